@@ -1,0 +1,82 @@
+"""BERT-style masked-language-model objective over MSA tokens.
+
+Capability parity: reference mlm.py:11-92.  `noise()` corrupts a
+(b, m, n) MSA during training (15% of maskable positions: mask token,
+keep-same, or random-replace); `forward()` computes the CE loss over the
+corrupted positions from the trunk's MSA embeddings.
+"""
+import math
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from . import constants
+
+
+def get_mask_subset_with_prob(mask, prob):
+    """Pick exactly ceil(prob * len) positions per row from `mask` (uniform
+    among allowed positions) — top-k on masked uniform noise."""
+    batch, seq_len = mask.shape
+    device = mask.device
+    max_masked = math.ceil(prob * seq_len)
+
+    num_tokens = mask.sum(dim=-1, keepdim=True)
+    # rows with fewer allowed positions than the global top-k get the
+    # excess samples suppressed
+    mask_excess = (mask.cumsum(dim=-1) > (num_tokens * prob).ceil())
+    mask_excess = mask_excess[:, :max_masked]
+
+    rand = torch.rand((batch, seq_len), device=device).masked_fill(~mask, -1e9)
+    _, sampled_indices = rand.topk(max_masked, dim=-1)
+    sampled_indices = (sampled_indices + 1).masked_fill_(mask_excess, 0)
+
+    new_mask = torch.zeros((batch, seq_len + 1), device=device)
+    new_mask.scatter_(-1, sampled_indices, 1)
+    return new_mask[:, 1:].bool()
+
+
+class MLM(nn.Module):
+    def __init__(self, dim, num_tokens, mask_id, mask_prob=0.15,
+                 random_replace_token_prob=0.1, keep_token_same_prob=0.1,
+                 exclude_token_ids=(0,)):
+        super().__init__()
+        self.to_logits = nn.Linear(dim, num_tokens)
+        self.mask_id = mask_id
+        self.mask_prob = mask_prob
+        self.exclude_token_ids = exclude_token_ids
+        self.keep_token_same_prob = keep_token_same_prob
+        self.random_replace_token_prob = random_replace_token_prob
+
+    def noise(self, seq, mask):
+        """Corrupt (b, m, n) MSA tokens; returns (noised, replaced_mask)."""
+        num_msa = seq.shape[1]
+        seq = seq.reshape(-1, seq.shape[-1])
+        mask = mask.reshape(-1, mask.shape[-1])
+
+        excluded_tokens_mask = mask
+        for token_id in self.exclude_token_ids:
+            excluded_tokens_mask = excluded_tokens_mask & (seq != token_id)
+
+        mlm_mask = get_mask_subset_with_prob(excluded_tokens_mask, self.mask_prob)
+
+        # of the selected positions: (1 - keep_same) get the mask token
+        seq = seq.masked_fill(mlm_mask, self.mask_id)
+
+        random_replace_mask = get_mask_subset_with_prob(
+            mlm_mask, (1 - self.keep_token_same_prob) * self.random_replace_token_prob)
+        random_tokens = torch.randint(1, constants.NUM_AMINO_ACIDS, seq.shape,
+                                      device=seq.device)
+        for token_id in self.exclude_token_ids:
+            random_replace_mask = random_replace_mask & (random_tokens != token_id)
+
+        noised_seq = torch.where(random_replace_mask, random_tokens, seq)
+        noised_seq = noised_seq.reshape(-1, num_msa, noised_seq.shape[-1])
+        mlm_mask = mlm_mask.reshape(-1, num_msa, mlm_mask.shape[-1])
+        return noised_seq, mlm_mask
+
+    def forward(self, seq_embed, original_seq, mask):
+        logits = self.to_logits(seq_embed)
+        seq_logits = logits[mask]
+        seq_labels = original_seq[mask]
+        return F.cross_entropy(seq_logits, seq_labels, reduction='mean')

@@ -1,0 +1,188 @@
+"""Invariant Point Attention — native implementation.
+
+Replaces the reference's external `invariant_point_attention` pip
+dependency (reference alphafold2.py:19, 608-615) with a from-scratch
+module implementing AF2's IPA (scalar qkv + frame-rotated point qkv +
+pair bias; attention over inter-point distances), keeping the same
+state-dict layout (`attn.to_scalar_q`, `attn.point_weights`,
+`attn.to_out`, `ff.{0,2,4}` ...) so checkpoints transfer.
+
+Runs in fp32 (the structure module is pinned to fp32 for equivariance —
+reference alphafold2.py:607,855).  This is the K7 fusion target of
+SURVEY.md §2.17: all einsums below are shaped for a single fused fp32
+HIP kernel per refinement iteration.
+"""
+import math
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from ..ops.fused_modules import FusedLayerNorm
+
+
+def exists(val):
+    return val is not None
+
+
+class InvariantPointAttention(nn.Module):
+    def __init__(self, *, dim, heads=8, scalar_key_dim=16, scalar_value_dim=16,
+                 point_key_dim=4, point_value_dim=4, pairwise_repr_dim=None,
+                 require_pairwise_repr=True, eps=1e-8):
+        super().__init__()
+        self.eps = eps
+        self.heads = heads
+        self.require_pairwise_repr = require_pairwise_repr
+
+        # three logit families (scalar / point / pair) share the softmax,
+        # each scaled so their variances match at init
+        num_attn_logits = 3 if require_pairwise_repr else 2
+
+        self.scalar_attn_logits_scale = (num_attn_logits * scalar_key_dim) ** -0.5
+        self.to_scalar_q = nn.Linear(dim, scalar_key_dim * heads, bias=False)
+        self.to_scalar_k = nn.Linear(dim, scalar_key_dim * heads, bias=False)
+        self.to_scalar_v = nn.Linear(dim, scalar_value_dim * heads, bias=False)
+
+        # learned per-head weight on the point-distance term, softplus-
+        # parameterized, init such that softplus(w) = 1
+        point_weight_init = math.log(math.exp(1.) - 1.)
+        self.point_weights = nn.Parameter(
+            torch.full((heads,), point_weight_init))
+        self.point_attn_logits_scale = (
+            (num_attn_logits * point_key_dim) * (9 / 2)) ** -0.5
+        self.to_point_q = nn.Linear(dim, point_key_dim * heads * 3, bias=False)
+        self.to_point_k = nn.Linear(dim, point_key_dim * heads * 3, bias=False)
+        self.to_point_v = nn.Linear(dim, point_value_dim * heads * 3, bias=False)
+
+        self.scalar_key_dim = scalar_key_dim
+        self.scalar_value_dim = scalar_value_dim
+        self.point_key_dim = point_key_dim
+        self.point_value_dim = point_value_dim
+
+        if require_pairwise_repr:
+            pairwise_repr_dim = pairwise_repr_dim if exists(pairwise_repr_dim) else dim
+            self.pairwise_attn_logits_scale = num_attn_logits ** -0.5
+            self.to_pairwise_attn_bias = nn.Sequential(
+                nn.Linear(pairwise_repr_dim, heads))
+        else:
+            pairwise_repr_dim = 0
+
+        self.to_out = nn.Linear(
+            heads * (scalar_value_dim + pairwise_repr_dim
+                     + point_value_dim * (3 + 1)), dim)
+
+    def forward(self, single_repr, pairwise_repr=None, *, rotations,
+                translations, mask=None):
+        x = single_repr
+        b, n, _ = x.shape
+        h = self.heads
+        assert not (self.require_pairwise_repr and not exists(pairwise_repr)), \
+            'pairwise representation must be given'
+
+        # scalar qkv -> (b, h, n, d)
+        def split_heads(t, d):
+            return t.reshape(b, n, h, d).permute(0, 2, 1, 3)
+
+        q_s = split_heads(self.to_scalar_q(x), self.scalar_key_dim)
+        k_s = split_heads(self.to_scalar_k(x), self.scalar_key_dim)
+        v_s = split_heads(self.to_scalar_v(x), self.scalar_value_dim)
+
+        # point qkv -> (b, h, n, p, 3), rotated/translated to global frame
+        def split_points(t, p):
+            return t.reshape(b, n, h, p, 3).permute(0, 2, 1, 3, 4)
+
+        q_p = split_points(self.to_point_q(x), self.point_key_dim)
+        k_p = split_points(self.to_point_k(x), self.point_key_dim)
+        v_p = split_points(self.to_point_v(x), self.point_value_dim)
+
+        # rotations (b, n, 3, 3); local -> global: R x + t
+        def to_global(p):
+            g = torch.einsum('b h n p c, b n c d -> b h n p d', p, rotations)
+            return g + translations[:, None, :, None, :]
+
+        q_p, k_p, v_p = map(to_global, (q_p, k_p, v_p))
+
+        # attention logits
+        attn_logits = torch.einsum('b h i d, b h j d -> b h i j', q_s, k_s) \
+            * self.scalar_attn_logits_scale
+
+        if self.require_pairwise_repr:
+            pair_bias = self.to_pairwise_attn_bias[0](pairwise_repr)  # (b,i,j,h)
+            attn_logits = attn_logits + pair_bias.permute(0, 3, 1, 2) \
+                * self.pairwise_attn_logits_scale
+
+        point_w = F.softplus(self.point_weights)  # (h,)
+        d2 = (q_p[:, :, :, None] - k_p[:, :, None, :]).pow(2).sum(dim=(-1, -2))
+        attn_logits = attn_logits - 0.5 * point_w[None, :, None, None] \
+            * self.point_attn_logits_scale * d2
+
+        if exists(mask):
+            mask_2d = mask[:, None, :, None] * mask[:, None, None, :]
+            mask_value = torch.finfo(attn_logits.dtype).max
+            attn_logits = attn_logits.masked_fill(~mask_2d, -mask_value)
+
+        attn = attn_logits.softmax(dim=-1)
+
+        # aggregate the three value families
+        out_scalar = torch.einsum('b h i j, b h j d -> b h i d', attn, v_s)
+        out_points_global = torch.einsum(
+            'b h i j, b h j p d -> b h i p d', attn, v_p)
+
+        # back to local frames: inverse of the row-vector convention
+        # g = l . R used in to_global, so l = g . R^T (contract g_d with R[c,d])
+        rel = out_points_global - translations[:, None, :, None, :]
+        out_points = torch.einsum('b h n p d, b n c d -> b h n p c',
+                                  rel, rotations)
+        out_points_norm = torch.sqrt(out_points.pow(2).sum(dim=-1) + self.eps)
+
+        pieces = [out_scalar.permute(0, 2, 1, 3).reshape(b, n, -1)]
+        if self.require_pairwise_repr:
+            out_pair = torch.einsum('b h i j, b i j d -> b h i d',
+                                    attn, pairwise_repr)
+            pieces.append(out_pair.permute(0, 2, 1, 3).reshape(b, n, -1))
+        pieces.append(out_points.permute(0, 2, 1, 3, 4).reshape(b, n, -1))
+        pieces.append(out_points_norm.permute(0, 2, 1, 3).reshape(b, n, -1))
+
+        return self.to_out(torch.cat(pieces, dim=-1))
+
+
+def _ipa_feedforward(dim, mult=1., num_layers=3, act=nn.ReLU):
+    layers = []
+    dim_hidden = int(dim * mult)
+    for ind in range(num_layers):
+        is_first = ind == 0
+        is_last = ind == (num_layers - 1)
+        dim_in = dim if is_first else dim_hidden
+        dim_out = dim if is_last else dim_hidden
+        layers.append(nn.Linear(dim_in, dim_out))
+        if not is_last:
+            layers.append(act())
+    return nn.Sequential(*layers)
+
+
+class IPABlock(nn.Module):
+    """IPA + transition feed-forward with post-norm residuals."""
+
+    def __init__(self, *, dim, ff_mult=1, ff_num_layers=3, post_norm=True,
+                 post_attn_dropout=0., post_ff_dropout=0., **kwargs):
+        super().__init__()
+        self.post_norm = post_norm
+        self.attn_norm = FusedLayerNorm(dim)
+        self.attn = InvariantPointAttention(dim=dim, **kwargs)
+        self.post_attn_dropout = nn.Dropout(post_attn_dropout)
+        self.ff_norm = FusedLayerNorm(dim)
+        self.ff = _ipa_feedforward(dim, mult=ff_mult, num_layers=ff_num_layers)
+        self.post_ff_dropout = nn.Dropout(post_ff_dropout)
+
+    def forward(self, x, **kwargs):
+        post_norm = self.post_norm
+        attn_input = x if post_norm else self.attn_norm(x)
+        x = self.attn(attn_input, **kwargs) + x
+        x = self.post_attn_dropout(x)
+        x = self.attn_norm(x) if post_norm else x
+
+        ff_input = x if post_norm else self.ff_norm(x)
+        x = self.ff(ff_input) + x
+        x = self.post_ff_dropout(x)
+        x = self.ff_norm(x) if post_norm else x
+        return x

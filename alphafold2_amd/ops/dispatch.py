@@ -1,0 +1,116 @@
+"""Backend dispatch between the gfx950 HIP extension and the eager path.
+
+Policy:
+* CPU tensors -> eager, always.
+* ROCm tensors + extension loaded -> HIP kernel when one exists for the
+  op (per-op hasattr check so kernels can land incrementally), else the
+  eager composition (which still runs on-device through rocBLAS/hipBLASLt).
+* ROCm tensors + extension NOT importable -> hard error ("native code not
+  loaded") unless AF2AMD_ALLOW_EAGER_GPU=1.
+* AF2AMD_FORCE_EAGER=1 pins eager everywhere (kernel parity tests).
+"""
+import os
+
+import torch
+
+from . import eager
+
+_FORCE_EAGER = os.environ.get("AF2AMD_FORCE_EAGER", "0") == "1"
+_ALLOW_EAGER_GPU = os.environ.get("AF2AMD_ALLOW_EAGER_GPU", "0") == "1"
+
+_EXT = None
+_EXT_TRIED = False
+_EXT_ERR = None
+
+
+def _load_ext():
+    global _EXT, _EXT_TRIED, _EXT_ERR
+    if _EXT_TRIED:
+        return _EXT
+    _EXT_TRIED = True
+    try:
+        from alphafold2_amd import _hip_ops  # built in-tree for gfx950
+        _EXT = _hip_ops
+    except ImportError as e:  # extension not built (CPU-only dev box)
+        _EXT = None
+        _EXT_ERR = e
+    return _EXT
+
+
+def hip_ops_available():
+    return _load_ext() is not None
+
+
+def _want_hip(t: torch.Tensor) -> bool:
+    if _FORCE_EAGER or not t.is_cuda:
+        return False
+    ext = _load_ext()
+    if ext is None:
+        if _ALLOW_EAGER_GPU:
+            return False
+        raise RuntimeError(
+            "alphafold2_amd._hip_ops is not importable on a GPU device "
+            f"(build with `python setup.py build_ext --inplace`): {_EXT_ERR}")
+    return True
+
+
+def using_hip(t: torch.Tensor, opname: str) -> bool:
+    """True when `opname` should run through the HIP extension."""
+    return _want_hip(t) and hasattr(_load_ext(), opname)
+
+
+# ---------------------------------------------------------------------------
+# op entry points
+
+
+def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
+                   tie_dim=None):
+    if using_hip(q, 'attn_fwd'):
+        from .hip_autograd import hip_attention_core
+        return hip_attention_core(q, k, v, bias=bias, mask=mask,
+                                  context_mask=context_mask, tie_dim=tie_dim)
+    return eager.attention_core(q, k, v, bias=bias, mask=mask,
+                                context_mask=context_mask, tie_dim=tie_dim)
+
+
+def geglu(x):
+    if using_hip(x, 'geglu_fwd'):
+        from .hip_autograd import hip_geglu
+        return hip_geglu(x)
+    return eager.geglu(x)
+
+
+def outer_product_mean(left, right, mask=None, eps=1e-5):
+    if using_hip(left, 'outer_mean_fwd'):
+        from .hip_autograd import hip_outer_product_mean
+        return hip_outer_product_mean(left, right, mask=mask, eps=eps)
+    return eager.outer_product_mean(left, right, mask=mask, eps=eps)
+
+
+def triangle_mix(left, right, mix):
+    if using_hip(left, 'trimix_fwd'):
+        from .hip_autograd import hip_triangle_mix
+        return hip_triangle_mix(left, right, mix)
+    return eager.triangle_mix(left, right, mix)
+
+
+def pair_outer_sum(x_left, x_right):
+    return eager.pair_outer_sum(x_left, x_right)
+
+
+def distance_buckets(coords, boundaries):
+    if using_hip(coords, 'dist_buckets'):
+        ext = _load_ext()
+        return ext.dist_buckets(coords.contiguous(), boundaries.contiguous())
+    return eager.distance_buckets(coords, boundaries)
+
+
+def layer_norm(x, weight, bias, eps=1e-5):
+    if using_hip(x, 'layernorm_fwd'):
+        from .hip_autograd import hip_layer_norm
+        return hip_layer_norm(x, weight, bias, eps)
+    return eager.layer_norm(x, weight, bias, eps)
+
+
+def softclamp_gate(x, gates):
+    return eager.softclamp_gate(x, gates)

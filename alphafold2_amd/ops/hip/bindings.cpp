@@ -1,0 +1,21 @@
+// Python bindings for the gfx950 HIP op library.
+#include <torch/extension.h>
+
+#include <vector>
+
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w,
+                                      at::Tensor b, double eps);
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
+                                      at::Tensor w, at::Tensor mean,
+                                      at::Tensor rstd);
+at::Tensor geglu_fwd(at::Tensor x);
+at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x);
+at::Tensor dist_buckets(at::Tensor coords, at::Tensor boundaries);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm forward (gfx950)");
+  m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm backward (gfx950)");
+  m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward (gfx950)");
+  m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward (gfx950)");
+  m.def("dist_buckets", &dist_buckets, "fused cdist+bucketize (gfx950)");
+}

@@ -1,0 +1,135 @@
+// Fused GEGLU activation — gfx950.
+//
+// FeedForward transition (reference alphafold2.py:69-94): the Linear
+// produces (..., 2H); GEGLU keeps a = x[..., :H], gates = x[..., H:],
+// out = a * gelu(gates).  Fusing the chunk + gelu + multiply saves two
+// full tensor read/writes on a memory-bound op (K6 epilogue of
+// SURVEY.md §2.17).  Vectorized 8-wide loads (guide G13).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+template <typename T, int VEC>
+__global__ void geglu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                 long rows, int H) {
+  const long total = rows * (long)H;
+  const long stride = (long)gridDim.x * blockDim.x * VEC;
+  for (long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
+       base < total; base += stride) {
+    const long row = base / H;
+    const int col = base - row * H;
+    const T* xr = x + row * (2L * H);
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) {
+      float a = to_f32(xr[col + k]);
+      float g = to_f32(xr[H + col + k]);
+      y[base + k] = from_f32<T>(a * gelu_f(g));
+    }
+  }
+}
+
+template <typename T, int VEC>
+__global__ void geglu_bwd_kernel(const T* __restrict__ dy,
+                                 const T* __restrict__ x,
+                                 T* __restrict__ dx, long rows, int H) {
+  const long total = rows * (long)H;
+  const long stride = (long)gridDim.x * blockDim.x * VEC;
+  for (long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
+       base < total; base += stride) {
+    const long row = base / H;
+    const int col = base - row * H;
+    const T* xr = x + row * (2L * H);
+    T* dxr = dx + row * (2L * H);
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) {
+      float a = to_f32(xr[col + k]);
+      float g = to_f32(xr[H + col + k]);
+      float go = to_f32(dy[base + k]);
+      dxr[col + k] = from_f32<T>(go * gelu_f(g));
+      dxr[H + col + k] = from_f32<T>(go * a * gelu_grad_f(g));
+    }
+  }
+}
+
+}  // namespace
+
+at::Tensor geglu_fwd(at::Tensor x) {
+  TORCH_CHECK(x.is_contiguous(), "geglu_fwd: x must be contiguous");
+  const int H2 = x.size(-1);
+  TORCH_CHECK(H2 % 2 == 0, "geglu_fwd: last dim must be even");
+  const int H = H2 / 2;
+  const long rows = x.numel() / H2;
+
+  auto sizes = x.sizes().vec();
+  sizes.back() = H;
+  auto y = at::empty(sizes, x.options());
+
+  const int block = 256;
+  const long total = rows * (long)H;
+  auto stream = at::cuda::getCurrentHIPStream();
+
+#define LAUNCH(T, VEC)                                                   \
+  do {                                                                   \
+    long grid = (total / VEC + block - 1) / block;                       \
+    if (grid > 2048) grid = 2048;                                        \
+    if (grid < 1) grid = 1;                                              \
+    hipLaunchKernelGGL((geglu_fwd_kernel<T, VEC>), dim3(grid),           \
+                       dim3(block), 0, stream,                           \
+                       reinterpret_cast<const T*>(x.data_ptr()),         \
+                       reinterpret_cast<T*>(y.data_ptr()), rows, H);     \
+  } while (0)
+
+  const bool vec8 = (H % 8) == 0;
+  if (x.scalar_type() == at::kBFloat16) {
+    if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
+  } else if (x.scalar_type() == at::kFloat) {
+    if (vec8) LAUNCH(float, 4); else LAUNCH(float, 1);
+  } else if (x.scalar_type() == at::kHalf) {
+    if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
+  } else {
+    TORCH_CHECK(false, "geglu_fwd: unsupported dtype");
+  }
+#undef LAUNCH
+  return y;
+}
+
+at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x) {
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous(),
+              "geglu_bwd: inputs must be contiguous");
+  const int H2 = x.size(-1);
+  const int H = H2 / 2;
+  const long rows = x.numel() / H2;
+  auto dx = at::empty_like(x);
+
+  const int block = 256;
+  const long total = rows * (long)H;
+  auto stream = at::cuda::getCurrentHIPStream();
+
+#define LAUNCH(T, VEC)                                                   \
+  do {                                                                   \
+    long grid = (total / VEC + block - 1) / block;                       \
+    if (grid > 2048) grid = 2048;                                        \
+    if (grid < 1) grid = 1;                                              \
+    hipLaunchKernelGGL((geglu_bwd_kernel<T, VEC>), dim3(grid),           \
+                       dim3(block), 0, stream,                           \
+                       reinterpret_cast<const T*>(dy.data_ptr()),        \
+                       reinterpret_cast<const T*>(x.data_ptr()),         \
+                       reinterpret_cast<T*>(dx.data_ptr()), rows, H);    \
+  } while (0)
+
+  const bool vec8 = (H % 8) == 0;
+  if (x.scalar_type() == at::kBFloat16) {
+    if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
+  } else if (x.scalar_type() == at::kFloat) {
+    if (vec8) LAUNCH(float, 4); else LAUNCH(float, 1);
+  } else if (x.scalar_type() == at::kHalf) {
+    if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
+  } else {
+    TORCH_CHECK(false, "geglu_bwd: unsupported dtype");
+  }
+#undef LAUNCH
+  return dx;
+}

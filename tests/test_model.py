@@ -1,0 +1,216 @@
+"""Model-level tests mirroring the reference suite
+(/root/reference/tests/test_attention.py) plus state-dict layout checks."""
+import torch
+
+from alphafold2_amd import Alphafold2
+
+
+def tiny_model(**kwargs):
+    defaults = dict(dim=32, depth=2, heads=2, dim_head=32)
+    defaults.update(kwargs)
+    return Alphafold2(**defaults)
+
+
+def test_main():
+    model = tiny_model()
+    seq = torch.randint(0, 21, (2, 64))
+    msa = torch.randint(0, 21, (2, 5, 64))
+    mask = torch.ones_like(seq).bool()
+    msa_mask = torch.ones_like(msa).bool()
+    ret = model(seq, msa, mask=mask, msa_mask=msa_mask)
+    assert ret.distance.shape == (2, 64, 64, 37)
+
+
+def test_no_msa():
+    model = tiny_model()
+    seq = torch.randint(0, 21, (2, 64))
+    mask = torch.ones_like(seq).bool()
+    ret = model(seq, mask=mask)
+    assert ret.distance.shape == (2, 64, 64, 37)
+
+
+def test_anglegrams():
+    model = tiny_model(predict_angles=True)
+    seq = torch.randint(0, 21, (2, 32))
+    msa = torch.randint(0, 21, (2, 5, 32))
+    mask = torch.ones_like(seq).bool()
+    msa_mask = torch.ones_like(msa).bool()
+    ret = model(seq, msa, mask=mask, msa_mask=msa_mask)
+    assert ret.theta_logits.shape == (2, 32, 32, 25)
+    assert ret.phi_logits.shape == (2, 32, 32, 13)
+    assert ret.omega_logits.shape == (2, 32, 32, 25)
+
+
+def test_templates():
+    model = tiny_model(templates_dim=32, templates_angles_feats_dim=32)
+    seq = torch.randint(0, 21, (2, 16))
+    mask = torch.ones_like(seq).bool()
+    msa = torch.randint(0, 21, (2, 5, 16))
+    msa_mask = torch.ones_like(msa).bool()
+    templates_feats = torch.randn(2, 3, 16, 16, 32)
+    templates_angles = torch.randn(2, 3, 16, 32)
+    templates_mask = torch.ones(2, 3, 16).bool()
+    ret = model(seq, msa, mask=mask, msa_mask=msa_mask,
+                templates_feats=templates_feats,
+                templates_angles=templates_angles,
+                templates_mask=templates_mask)
+    assert ret.distance.shape == (2, 16, 16, 37)
+
+
+def test_extra_msa():
+    model = tiny_model(dim=64, predict_coords=True)
+    seq = torch.randint(0, 21, (2, 4))
+    mask = torch.ones_like(seq).bool()
+    msa = torch.randint(0, 21, (2, 5, 4))
+    msa_mask = torch.ones_like(msa).bool()
+    extra_msa = torch.randint(0, 21, (2, 5, 4))
+    extra_msa_mask = torch.ones_like(extra_msa).bool()
+    coords = model(seq, msa, mask=mask, msa_mask=msa_mask,
+                   extra_msa=extra_msa, extra_msa_mask=extra_msa_mask)
+    assert coords.shape == (2, 4, 3)
+
+
+def test_embeddings():
+    model = tiny_model()
+    seq = torch.randint(0, 21, (2, 16))
+    mask = torch.ones_like(seq).bool()
+    embedds = torch.randn(2, 1, 16, 1280)
+
+    ret = model(seq, mask=mask, embedds=embedds, msa_mask=None)
+    assert ret.distance.shape == (2, 16, 16, 37)
+
+    embedds_mask = torch.ones_like(embedds[..., -1]).bool()
+    ret = model(seq, mask=mask, embedds=embedds, msa_mask=embedds_mask)
+    assert ret.distance.shape == (2, 16, 16, 37)
+
+
+def test_coords():
+    model = tiny_model(predict_coords=True, structure_module_depth=1,
+                       structure_module_heads=1, structure_module_dim_head=1)
+    seq = torch.randint(0, 21, (2, 16))
+    mask = torch.ones_like(seq).bool()
+    msa = torch.randint(0, 21, (2, 5, 16))
+    msa_mask = torch.ones_like(msa).bool()
+    coords = model(seq, msa, mask=mask, msa_mask=msa_mask)
+    assert coords.shape == (2, 16, 3), 'must output coordinates'
+
+
+def test_coords_backwards():
+    model = tiny_model(dim=64, predict_coords=True,
+                       structure_module_depth=1, structure_module_heads=1,
+                       structure_module_dim_head=1)
+    seq = torch.randint(0, 21, (2, 16))
+    mask = torch.ones_like(seq).bool()
+    msa = torch.randint(0, 21, (2, 5, 16))
+    msa_mask = torch.ones_like(msa).bool()
+    coords = model(seq, msa, mask=mask, msa_mask=msa_mask)
+    coords.sum().backward()
+    grads = [p.grad for p in model.parameters() if p.grad is not None]
+    assert len(grads) > 0, 'gradients must flow back through the structure module'
+
+
+def test_confidence():
+    model = tiny_model(dim=64, depth=1, predict_coords=True)
+    seq = torch.randint(0, 21, (2, 16))
+    mask = torch.ones_like(seq).bool()
+    msa = torch.randint(0, 21, (2, 5, 16))
+    msa_mask = torch.ones_like(msa).bool()
+    coords, confidences = model(seq, msa, mask=mask, msa_mask=msa_mask,
+                                return_confidence=True)
+    assert coords.shape[:-1] == confidences.shape[:-1]
+
+
+def test_recycling():
+    model = tiny_model(dim=64, predict_coords=True)
+    seq = torch.randint(0, 21, (2, 4))
+    mask = torch.ones_like(seq).bool()
+    msa = torch.randint(0, 21, (2, 5, 4))
+    msa_mask = torch.ones_like(msa).bool()
+    extra_msa = torch.randint(0, 21, (2, 5, 4))
+    extra_msa_mask = torch.ones_like(extra_msa).bool()
+
+    coords, ret = model(seq, msa, mask=mask, msa_mask=msa_mask,
+                        extra_msa=extra_msa, extra_msa_mask=extra_msa_mask,
+                        return_aux_logits=True, return_recyclables=True)
+    coords, ret = model(seq, msa, mask=mask, msa_mask=msa_mask,
+                        extra_msa=extra_msa, extra_msa_mask=extra_msa_mask,
+                        recyclables=ret.recyclables,
+                        return_aux_logits=True, return_recyclables=True)
+    assert coords.shape == (2, 4, 3)
+
+
+def test_mlm_loss_during_training():
+    model = tiny_model()
+    model.train()
+    seq = torch.randint(0, 21, (2, 32))
+    msa = torch.randint(0, 21, (2, 5, 32))
+    mask = torch.ones_like(seq).bool()
+    msa_mask = torch.ones_like(msa).bool()
+    ret = model(seq, msa, mask=mask, msa_mask=msa_mask)
+    assert ret.msa_mlm_loss is not None
+    assert ret.msa_mlm_loss.requires_grad
+
+
+def test_state_dict_layout():
+    """Checkpoint layout parity with the reference model
+    (reference alphafold2.py:469-628 module attribute map)."""
+    model = tiny_model(predict_angles=True, predict_coords=True)
+    keys = set(model.state_dict().keys())
+
+    expected = [
+        'token_emb.weight',
+        'to_pairwise_repr.weight',
+        'pos_emb.weight',
+        'to_template_embed.weight',
+        'template_angle_mlp.0.weight',
+        'template_angle_mlp.2.weight',
+        'to_prob_theta.weight',
+        'to_prob_phi.weight',
+        'to_prob_omega.weight',
+        'embedd_project.weight',
+        'mlm.to_logits.weight',
+        'to_distogram_logits.0.weight',
+        'to_distogram_logits.1.weight',
+        'msa_to_single_repr_dim.weight',
+        'trunk_to_pairwise_repr_dim.weight',
+        'ipa_block.attn.to_scalar_q.weight',
+        'ipa_block.attn.point_weights',
+        'ipa_block.attn.to_out.weight',
+        'ipa_block.ff.0.weight',
+        'ipa_block.ff.2.weight',
+        'ipa_block.ff.4.weight',
+        'to_quaternion_update.weight',
+        'to_points.weight',
+        'lddt_linear.weight',
+        'recycling_msa_norm.weight',
+        'recycling_pairwise_norm.weight',
+        'recycling_distance_embed.weight',
+        # trunk block 0 internals
+        'net.layers.0.layer.0.outer_mean.left_proj.weight',
+        'net.layers.0.layer.0.triangle_multiply_outgoing.left_proj.weight',
+        'net.layers.0.layer.0.triangle_attention_outgoing.attn.to_q.weight',
+        'net.layers.0.layer.0.triangle_attention_outgoing.edges_to_attn_bias.0.weight',
+        'net.layers.0.layer.1.net.0.weight',
+        'net.layers.0.layer.1.net.3.weight',
+        'net.layers.0.layer.2.row_attn.attn.gating.weight',
+        'net.layers.0.layer.3.norm.weight',
+        'extra_msa_evoformer.layers.0.layer.0.outer_mean.norm.weight',
+        'template_pairwise_embedder.triangle_multiply_ingoing.out_gate.weight',
+        'template_pointwise_attn.to_kv.weight',
+    ]
+    missing = [k for k in expected if k not in keys]
+    assert not missing, f'missing reference-layout keys: {missing}'
+
+
+def test_reversible_trunk():
+    model = tiny_model(dim=32, depth=2, reversible=True)
+    model.train()
+    seq = torch.randint(0, 21, (2, 16))
+    msa = torch.randint(0, 21, (2, 3, 16))
+    mask = torch.ones_like(seq).bool()
+    msa_mask = torch.ones_like(msa).bool()
+    ret = model(seq, msa, mask=mask, msa_mask=msa_mask)
+    loss = ret.distance.sum() + ret.msa_mlm_loss
+    loss.backward()
+    grads = [p.grad for p in model.net.parameters() if p.grad is not None]
+    assert len(grads) > 0
