@@ -102,7 +102,7 @@ class Attention(nn.Module):
         init_zero_(self.to_out)
 
     def forward(self, x, mask=None, attn_bias=None, context=None,
-                context_mask=None, tie_dim=None):
+                context_mask=None, tie_dim=None, attn_bias_repeat=1):
         h = self.heads
         has_context = exists(context)
         context = default(context, x)
@@ -121,7 +121,7 @@ class Attention(nn.Module):
 
         out = ops.attention_core(
             q, k, v, bias=attn_bias, mask=mask, context_mask=context_mask,
-            tie_dim=tie_dim)
+            tie_dim=tie_dim, bias_repeat=attn_bias_repeat)
 
         out = out.transpose(-2, -3).reshape(*x.shape[:-1], -1)
 
@@ -170,15 +170,19 @@ class AxialAttention(nn.Module):
             inp = x.reshape(b * h, w, d)
             m = mask.reshape(b * h, w) if exists(mask) else None
 
+        # pair-rep bias: kept at (b, heads, i, j); the fold over the
+        # axial dim is a broadcast handled inside the attention core
+        # (the eager reference materializes (b*axial, h, i, j) —
+        # reference alphafold2.py:248)
         attn_bias = None
         if exists(self.edges_to_attn_bias) and exists(edges):
             bias = self.edges_to_attn_bias[0](edges)         # (b, i, j, heads)
-            bias = bias.permute(0, 3, 1, 2)                  # (b, heads, i, j)
-            attn_bias = bias.repeat_interleave(axial_dim, dim=0)
+            attn_bias = bias.permute(0, 3, 1, 2).contiguous()
 
         tie_dim = axial_dim if self.global_query_attn else None
 
-        out = self.attn(inp, mask=m, attn_bias=attn_bias, tie_dim=tie_dim)
+        out = self.attn(inp, mask=m, attn_bias=attn_bias, tie_dim=tie_dim,
+                        attn_bias_repeat=axial_dim if exists(attn_bias) else 1)
 
         if self.col_attn:
             out = out.reshape(b, w, h, d).permute(0, 2, 1, 3)

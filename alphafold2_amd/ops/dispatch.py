@@ -64,11 +64,24 @@ def using_hip(t: torch.Tensor, opname: str) -> bool:
 
 
 def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
-                   tie_dim=None):
-    if using_hip(q, 'attn_fwd'):
+                   tie_dim=None, bias_repeat=1):
+    """bias (when given) has shape (B // bias_repeat, h, Lq, Lk); the
+    repeat fold is resolved inside the fused kernel (never materialized),
+    and expanded explicitly only on the eager path."""
+    fusable = (
+        tie_dim is None
+        and q.dtype == torch.bfloat16
+        and q.shape[-1] == 64
+        and (bias is None or bias.dtype == torch.bfloat16)
+        and using_hip(q, 'attn_fwd')
+    )
+    if fusable:
         from .hip_autograd import hip_attention_core
         return hip_attention_core(q, k, v, bias=bias, mask=mask,
-                                  context_mask=context_mask, tie_dim=tie_dim)
+                                  context_mask=context_mask,
+                                  bias_repeat=bias_repeat)
+    if bias is not None and bias_repeat != 1:
+        bias = bias.repeat_interleave(bias_repeat, dim=0)
     return eager.attention_core(q, k, v, bias=bias, mask=mask,
                                 context_mask=context_mask, tie_dim=tie_dim)
 

@@ -11,6 +11,16 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
 at::Tensor geglu_fwd(at::Tensor x);
 at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x);
 at::Tensor dist_buckets(at::Tensor coords, at::Tensor boundaries);
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 c10::optional<at::Tensor> bias,
+                                 c10::optional<at::Tensor> mask,
+                                 long bias_repeat, double scale);
+std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
+                                 at::Tensor v, at::Tensor out, at::Tensor lse,
+                                 c10::optional<at::Tensor> bias,
+                                 c10::optional<at::Tensor> mask,
+                                 long bias_repeat, double scale,
+                                 bool need_dbias);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm forward (gfx950)");
@@ -18,4 +28,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward (gfx950)");
   m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward (gfx950)");
   m.def("dist_buckets", &dist_buckets, "fused cdist+bucketize (gfx950)");
+  m.def("attn_fwd", &attn_fwd, "fused flash attention forward (gfx950)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
+        py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"));
+  m.def("attn_bwd", &attn_bwd, "fused flash attention backward (gfx950)");
 }

@@ -44,10 +44,52 @@ def hip_geglu(x):
     return _GegluFn.apply(x)
 
 
-# placeholders — wired when the corresponding kernels land
+class _AttentionFn(torch.autograd.Function):
+    """Fused flash attention (bf16, head dim 64) with broadcast pair bias.
+
+    bias has shape (B // bias_repeat, h, Lq, Lk); consecutive groups of
+    `bias_repeat` batch entries share a bias slice (the axial-attention
+    fold — never materialized).
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias, mask, bias_repeat, scale):
+        ext = _load_ext()
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        bias_c = bias.contiguous() if bias is not None else None
+        mask_c = mask.contiguous() if mask is not None else None
+        out, lse = ext.attn_fwd(q, k, v, bias_c, mask_c, bias_repeat, scale)
+        ctx.save_for_backward(q, k, v, out, lse,
+                              *( [bias_c] if bias_c is not None else [] ))
+        ctx.has_bias = bias_c is not None
+        ctx.mask = mask_c
+        ctx.bias_repeat = bias_repeat
+        ctx.scale = scale
+        ctx.bias_requires_grad = bias is not None and bias.requires_grad
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _load_ext()
+        saved = ctx.saved_tensors
+        q, k, v, out, lse = saved[:5]
+        bias = saved[5] if ctx.has_bias else None
+        need_dbias = ctx.bias_requires_grad
+        rets = ext.attn_bwd(dout.contiguous(), q, k, v, out, lse, bias,
+                            ctx.mask, ctx.bias_repeat, ctx.scale, need_dbias)
+        dq, dk, dv = rets[:3]
+        dbias = rets[3].to(bias.dtype) if need_dbias else None
+        return dq, dk, dv, dbias, None, None, None
+
+
 def hip_attention_core(q, k, v, bias=None, mask=None, context_mask=None,
-                       tie_dim=None):
-    raise NotImplementedError
+                       tie_dim=None, bias_repeat=1):
+    assert tie_dim is None, "tie_dim handled by the eager path"
+    key_mask = context_mask if context_mask is not None else mask
+    if key_mask is not None:
+        key_mask = key_mask.to(torch.uint8)
+    scale = q.shape[-1] ** -0.5
+    return _AttentionFn.apply(q, k, v, bias, key_mask, bias_repeat, scale)
 
 
 def hip_outer_product_mean(left, right, mask=None, eps=1e-5):

@@ -22,6 +22,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "layernorm.hip"),
         os.path.join(HIP_DIR, "geglu.hip"),
         os.path.join(HIP_DIR, "distbucket.hip"),
+        os.path.join(HIP_DIR, "attention.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
