@@ -650,16 +650,18 @@ void attn_bwd_dkv_kernel(const bf16_t* __restrict__ q,
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int kvrow = wave * 16 + (lane >> 4) * 4 + reg;
-        float ds = pt[c][reg] * (dpt[c][reg] - dl) * scale;
+        // dL/d(S*scale + bias) — the bias gradient has NO scale factor;
+        // scale applies only on the path back into Q K^T
+        float ds_total = pt[c][reg] * (dpt[c][reg] - dl);
         if (NEED_DBIAS && qcol < q_rows && kvrow < kv_rows) {
-          // fold over bias_repeat with fp32 atomics (dbias /= nothing:
-          // sum over the repeat group is the correct gradient)
+          // fold over bias_repeat with fp32 atomics (sum over the
+          // repeat group is the correct gradient)
           atomicAdd(&dbias_g[(long)(t * BQ + qcol) * Lk
-                             + ktile * BK + kvrow], ds);
+                             + ktile * BK + kvrow], ds_total);
         }
         const int row = (lane >> 4) * 4 + reg;
         *reinterpret_cast<bf16_t*>(sw + swz(row, qcol * (int)sizeof(bf16_t)))
-            = (bf16_t)ds;
+            = (bf16_t)(ds_total * scale);
       }
     }
     __builtin_amdgcn_s_waitcnt(0);

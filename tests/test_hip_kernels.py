@@ -15,7 +15,7 @@ def ext():
 
 
 @pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-5),
-                                       (torch.bfloat16, 2e-2)])
+                                       (torch.bfloat16, 5e-2)])
 @pytest.mark.parametrize("shape", [(128, 256), (64, 64, 384), (7, 33)])
 def test_layernorm_parity(ext, dtype, tol, shape):
     from alphafold2_amd.ops.hip_autograd import hip_layer_norm
