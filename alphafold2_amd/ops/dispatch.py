@@ -126,4 +126,7 @@ def layer_norm(x, weight, bias, eps=1e-5):
 
 
 def softclamp_gate(x, gates):
+    if x.shape == gates.shape and using_hip(x, 'gatemul_fwd'):
+        from .hip_autograd import hip_gatemul
+        return hip_gatemul(x, gates)
     return eager.softclamp_gate(x, gates)

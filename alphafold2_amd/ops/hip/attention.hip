@@ -51,10 +51,30 @@ __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return row * ROWB + (byte_in_row ^ ((row & 7) << 4));
 }
 
+// strided (B, h, L, 64) tensor view: element strides, innermost dim
+// contiguous.  Lets the kernels consume the Linear outputs' natural
+// (B, L, h*64) layout through a permuted view — no contiguous() copies.
+struct TView {
+  const bf16_t* p;
+  long bs, hs, rs;  // batch, head, row strides (elements)
+  __device__ const bf16_t* base(int b, int h) const {
+    return p + (long)b * bs + (long)h * hs;
+  }
+};
+struct TViewMut {
+  bf16_t* p;
+  long bs, hs, rs;
+  __device__ bf16_t* base(int b, int h) const {
+    return p + (long)b * bs + (long)h * hs;
+  }
+};
+
 // cooperative stage of a [rows<=64][64] bf16 tile global->LDS (swizzled).
 // Each of 256 threads moves 2 16-byte chunks.  OOB rows zero-filled.
+// row_stride in elements (bf16); rows are 64-element contiguous.
 __device__ __forceinline__ void stage_tile(const bf16_t* __restrict__ g,
-                                           int rows, char* lds) {
+                                           long row_stride, int rows,
+                                           char* lds) {
   const int tid = threadIdx.x;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
@@ -64,7 +84,7 @@ __device__ __forceinline__ void stage_tile(const bf16_t* __restrict__ g,
     float4 val = {0, 0, 0, 0};
     if (row < rows) {
       val = *reinterpret_cast<const float4*>(
-          reinterpret_cast<const char*>(g) + row * ROWB + c16);
+          reinterpret_cast<const char*>(g + row * row_stride) + c16);
     }
     *reinterpret_cast<float4*>(lds + swz(row, c16)) = val;
   }
@@ -116,12 +136,10 @@ __device__ __forceinline__ float rowred_sum(const f32x4 s[4], int reg) {
 
 template <bool HAS_BIAS, bool HAS_MASK>
 __global__ __launch_bounds__(256, 2)
-void attn_fwd_kernel(const bf16_t* __restrict__ q,
-                     const bf16_t* __restrict__ k,
-                     const bf16_t* __restrict__ v,
+void attn_fwd_kernel(TView q, TView k, TView v,
                      const bf16_t* __restrict__ bias,
                      const unsigned char* __restrict__ mask,
-                     bf16_t* __restrict__ out, float* __restrict__ lse,
+                     TViewMut out, float* __restrict__ lse,
                      int Lq, int Lk, int heads, int bias_repeat,
                      float scale) {
   __shared__ char q_lds[BQ * ROWB];
@@ -137,11 +155,9 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  const long qk_stride = (long)Lq * DH;
-  const long kv_stride = (long)Lk * DH;
-  const bf16_t* q_g = q + (long)bh * qk_stride + (long)qtile * BQ * DH;
-  const bf16_t* k_g = k + (long)bh * kv_stride;
-  const bf16_t* v_g = v + (long)bh * kv_stride;
+  const bf16_t* q_g = q.base(batch, head) + (long)qtile * BQ * q.rs;
+  const bf16_t* k_g = k.base(batch, head);
+  const bf16_t* v_g = v.base(batch, head);
   const bf16_t* bias_g = nullptr;
   if (HAS_BIAS) {
     const int bias_batch = batch / bias_repeat;
@@ -150,7 +166,7 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q,
   }
 
   const int q_rows = min(BQ, Lq - qtile * BQ);
-  stage_tile(q_g, q_rows, q_lds);
+  stage_tile(q_g, q.rs, q_rows, q_lds);
   __syncthreads();
 
   // per-wave Q fragments (rows wave*16 + (lane&15))
@@ -173,8 +189,8 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q,
   for (int t = 0; t < n_kv; ++t) {
     const int kv_rows = min(BK, Lk - t * BK);
     __syncthreads();
-    stage_tile(k_g + (long)t * BK * DH, kv_rows, k_lds);
-    stage_tile(v_g + (long)t * BK * DH, kv_rows, v_lds);
+    stage_tile(k_g + (long)t * BK * k.rs, k.rs, kv_rows, k_lds);
+    stage_tile(v_g + (long)t * BK * v.rs, v.rs, kv_rows, v_lds);
     if (HAS_MASK && threadIdx.x < BK) {
       m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
           ? mask[(long)batch * Lk + t * BK + threadIdx.x] : 0;
@@ -266,7 +282,7 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q,
   }
 
   // epilogue: O /= l, store out + lse
-  bf16_t* out_g = out + (long)bh * qk_stride + (long)qtile * BQ * DH;
+  bf16_t* out_g = out.base(batch, head) + (long)qtile * BQ * out.rs;
   float* lse_g = lse + (long)bh * Lq + (long)qtile * BQ;
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
@@ -275,7 +291,7 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q,
     if (row < q_rows) {
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
-        out_g[(long)row * DH + c * 16 + (lane & 15)] =
+        out_g[(long)row * out.rs + c * 16 + (lane & 15)] =
             (bf16_t)(o_acc[c][reg] * linv);
       }
       if ((lane & 15) == 0) {
@@ -288,14 +304,18 @@ void attn_fwd_kernel(const bf16_t* __restrict__ q,
 // ---------------------------------------------------------------------------
 // backward: delta = rowsum(dO * O)
 
-__global__ void attn_delta_kernel(const bf16_t* __restrict__ dout,
-                                  const bf16_t* __restrict__ out,
-                                  float* __restrict__ delta, long rows) {
-  // one 64-lane wave per row of DH=64
+__global__ void attn_delta_kernel(TView dout, TView out,
+                                  float* __restrict__ delta,
+                                  int heads, int Lq, long rows) {
+  // one 64-lane wave per (b, h, l) row of DH=64
   const long row = (long)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
   if (row >= rows) return;
   const int lane = threadIdx.x & 63;
-  float v = to_f32(dout[row * DH + lane]) * to_f32(out[row * DH + lane]);
+  const int l = row % Lq;
+  const int bh = row / Lq;
+  const int b = bh / heads, h = bh - (long)(bh / heads) * heads;
+  float v = to_f32(dout.base(b, h)[(long)l * dout.rs + lane]) *
+            to_f32(out.base(b, h)[(long)l * out.rs + lane]);
   v = wave_reduce_sum(v);
   if (lane == 0) delta[row] = v;
 }
@@ -305,15 +325,13 @@ __global__ void attn_delta_kernel(const bf16_t* __restrict__ dout,
 
 template <bool HAS_BIAS, bool HAS_MASK>
 __global__ __launch_bounds__(256, 2)
-void attn_bwd_dq_kernel(const bf16_t* __restrict__ q,
-                        const bf16_t* __restrict__ k,
-                        const bf16_t* __restrict__ v,
+void attn_bwd_dq_kernel(TView q, TView k, TView v,
                         const bf16_t* __restrict__ bias,
                         const unsigned char* __restrict__ mask,
-                        const bf16_t* __restrict__ dout,
+                        TView dout,
                         const float* __restrict__ lse,
                         const float* __restrict__ delta,
-                        bf16_t* __restrict__ dq,
+                        TViewMut dq,
                         int Lq, int Lk, int heads, int bias_repeat,
                         float scale) {
   __shared__ char q_lds[BQ * ROWB];
@@ -330,12 +348,10 @@ void attn_bwd_dq_kernel(const bf16_t* __restrict__ q,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  const long qk_stride = (long)Lq * DH;
-  const long kv_stride = (long)Lk * DH;
-  const bf16_t* q_g = q + (long)bh * qk_stride + (long)qtile * BQ * DH;
-  const bf16_t* do_g = dout + (long)bh * qk_stride + (long)qtile * BQ * DH;
-  const bf16_t* k_g = k + (long)bh * kv_stride;
-  const bf16_t* v_g = v + (long)bh * kv_stride;
+  const bf16_t* q_g = q.base(batch, head) + (long)qtile * BQ * q.rs;
+  const bf16_t* do_g = dout.base(batch, head) + (long)qtile * BQ * dout.rs;
+  const bf16_t* k_g = k.base(batch, head);
+  const bf16_t* v_g = v.base(batch, head);
   const float* lse_g = lse + (long)bh * Lq + (long)qtile * BQ;
   const float* delta_g = delta + (long)bh * Lq + (long)qtile * BQ;
   const bf16_t* bias_g = nullptr;
@@ -346,8 +362,8 @@ void attn_bwd_dq_kernel(const bf16_t* __restrict__ q,
   }
 
   const int q_rows = min(BQ, Lq - qtile * BQ);
-  stage_tile(q_g, q_rows, q_lds);
-  stage_tile(do_g, q_rows, do_lds);
+  stage_tile(q_g, q.rs, q_rows, q_lds);
+  stage_tile(do_g, dout.rs, q_rows, do_lds);
   __syncthreads();
 
   const int qrow = wave * 16 + (lane & 15);
@@ -375,8 +391,8 @@ void attn_bwd_dq_kernel(const bf16_t* __restrict__ q,
   for (int t = 0; t < n_kv; ++t) {
     const int kv_rows = min(BK, Lk - t * BK);
     __syncthreads();
-    stage_tile(k_g + (long)t * BK * DH, kv_rows, k_lds);
-    stage_tile(v_g + (long)t * BK * DH, kv_rows, v_lds);
+    stage_tile(k_g + (long)t * BK * k.rs, k.rs, kv_rows, k_lds);
+    stage_tile(v_g + (long)t * BK * v.rs, v.rs, kv_rows, v_lds);
     if (HAS_MASK && threadIdx.x < BK) {
       m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
           ? mask[(long)batch * Lk + t * BK + threadIdx.x] : 0;
@@ -458,14 +474,14 @@ void attn_bwd_dq_kernel(const bf16_t* __restrict__ q,
     }
   }
 
-  bf16_t* dq_g = dq + (long)bh * qk_stride + (long)qtile * BQ * DH;
+  bf16_t* dq_g = dq.base(batch, head) + (long)qtile * BQ * dq.rs;
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
     const int row = wave * 16 + (lane >> 4) * 4 + reg;
     if (row < q_rows) {
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
-        dq_g[(long)row * DH + c * 16 + (lane & 15)] =
+        dq_g[(long)row * dq.rs + c * 16 + (lane & 15)] =
             (bf16_t)dq_acc[c][reg];
       }
     }
@@ -480,15 +496,13 @@ void attn_bwd_dq_kernel(const bf16_t* __restrict__ q,
 
 template <bool HAS_BIAS, bool HAS_MASK, bool NEED_DBIAS>
 __global__ __launch_bounds__(256, 2)
-void attn_bwd_dkv_kernel(const bf16_t* __restrict__ q,
-                         const bf16_t* __restrict__ k,
-                         const bf16_t* __restrict__ v,
+void attn_bwd_dkv_kernel(TView q, TView k, TView v,
                          const bf16_t* __restrict__ bias,
                          const unsigned char* __restrict__ mask,
-                         const bf16_t* __restrict__ dout,
+                         TView dout,
                          const float* __restrict__ lse,
                          const float* __restrict__ delta,
-                         bf16_t* __restrict__ dk, bf16_t* __restrict__ dv,
+                         TViewMut dk, TViewMut dv,
                          float* __restrict__ dbias,
                          int Lq, int Lk, int heads, int bias_repeat,
                          float scale) {
@@ -508,12 +522,10 @@ void attn_bwd_dkv_kernel(const bf16_t* __restrict__ q,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  const long qk_stride = (long)Lq * DH;
-  const long kv_stride = (long)Lk * DH;
-  const bf16_t* k_g = k + (long)bh * kv_stride + (long)ktile * BK * DH;
-  const bf16_t* v_g = v + (long)bh * kv_stride + (long)ktile * BK * DH;
-  const bf16_t* q_g = q + (long)bh * qk_stride;
-  const bf16_t* do_g = dout + (long)bh * qk_stride;
+  const bf16_t* k_g = k.base(batch, head) + (long)ktile * BK * k.rs;
+  const bf16_t* v_g = v.base(batch, head) + (long)ktile * BK * v.rs;
+  const bf16_t* q_g = q.base(batch, head);
+  const bf16_t* do_g = dout.base(batch, head);
   const float* lse_g = lse + (long)bh * Lq;
   const float* delta_g = delta + (long)bh * Lq;
   const bf16_t* bias_g = nullptr;
@@ -525,8 +537,8 @@ void attn_bwd_dkv_kernel(const bf16_t* __restrict__ q,
     dbias_g = dbias + (long)(bias_batch * heads + head) * Lq * Lk;
 
   const int kv_rows = min(BK, Lk - ktile * BK);
-  stage_tile(k_g, kv_rows, k_lds);
-  stage_tile(v_g, kv_rows, v_lds);
+  stage_tile(k_g, k.rs, kv_rows, k_lds);
+  stage_tile(v_g, v.rs, kv_rows, v_lds);
   if (HAS_MASK && threadIdx.x < BK) {
     m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
         ? mask[(long)batch * Lk + ktile * BK + threadIdx.x] : 0;
@@ -561,8 +573,8 @@ void attn_bwd_dkv_kernel(const bf16_t* __restrict__ q,
   for (int t = 0; t < n_q; ++t) {
     const int q_rows = min(BQ, Lq - t * BQ);
     __syncthreads();
-    stage_tile(q_g + (long)t * BQ * DH, q_rows, q_lds);
-    stage_tile(do_g + (long)t * BQ * DH, q_rows, do_lds);
+    stage_tile(q_g + (long)t * BQ * q.rs, q.rs, q_rows, q_lds);
+    stage_tile(do_g + (long)t * BQ * dout.rs, dout.rs, q_rows, do_lds);
     if (threadIdx.x < BQ) {
       const int qq = t * BQ + threadIdx.x;
       lse_lds[threadIdx.x] = (threadIdx.x < q_rows) ? lse_g[qq] : NEG_INF;
@@ -684,17 +696,17 @@ void attn_bwd_dkv_kernel(const bf16_t* __restrict__ q,
     }
   }
 
-  bf16_t* dk_g = dk + (long)bh * kv_stride + (long)ktile * BK * DH;
-  bf16_t* dv_g = dv + (long)bh * kv_stride + (long)ktile * BK * DH;
+  bf16_t* dk_g = dk.base(batch, head) + (long)ktile * BK * dk.rs;
+  bf16_t* dv_g = dv.base(batch, head) + (long)ktile * BK * dv.rs;
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
     const int row = wave * 16 + (lane >> 4) * 4 + reg;
     if (row < kv_rows) {
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
-        dk_g[(long)row * DH + c * 16 + (lane & 15)] =
+        dk_g[(long)row * dk.rs + c * 16 + (lane & 15)] =
             (bf16_t)dk_acc[c][reg];
-        dv_g[(long)row * DH + c * 16 + (lane & 15)] =
+        dv_g[(long)row * dv.rs + c * 16 + (lane & 15)] =
             (bf16_t)dv_acc[c][reg];
       }
     }
@@ -706,34 +718,54 @@ void attn_bwd_dkv_kernel(const bf16_t* __restrict__ q,
 // ---------------------------------------------------------------------------
 // host launchers
 
+namespace {
+
+TView make_view(const at::Tensor& t) {
+  TORCH_CHECK(t.dim() == 4 && t.size(3) == DH && t.stride(3) == 1,
+              "attention tensors must be 4-D (B, h, L, 64), last dim dense");
+  TORCH_CHECK(t.stride(2) % 8 == 0 && t.stride(1) % 8 == 0,
+              "attention tensor strides must be 16-byte aligned");
+  return TView{reinterpret_cast<const bf16_t*>(t.data_ptr()),
+               t.stride(0), t.stride(1), t.stride(2)};
+}
+
+TViewMut make_view_mut(at::Tensor& t) {
+  TView v = make_view(t);
+  return TViewMut{const_cast<bf16_t*>(v.p), v.bs, v.hs, v.rs};
+}
+
+}  // namespace
+
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
                                  long bias_repeat, double scale) {
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
-  TORCH_CHECK(q.size(-1) == DH, "attn_fwd: head dim must be 64");
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
 
-  auto out = at::empty_like(q);
+  // out in (B, Lq, H, DH) memory (the layout downstream Linears want),
+  // exposed as a (B, H, Lq, DH) strided view
+  auto out_bnhd = at::empty({B, Lq, H, DH}, q.options());
+  auto out = out_bnhd.permute({0, 2, 1, 3});
   auto lse = at::empty({B, H, Lq}, q.options().dtype(at::kFloat));
 
   const bool has_bias = bias.has_value();
   const bool has_mask = mask.has_value();
-  at::Tensor mask_u8;
+  at::Tensor bias_c, mask_u8;
+  if (has_bias) bias_c = bias->contiguous();
   if (has_mask) mask_u8 = mask->to(at::kByte).contiguous();
 
   dim3 grid((Lq + BQ - 1) / BQ, B * H);
   auto stream = at::cuda::getCurrentHIPStream();
+  TView qv = make_view(q), kv = make_view(k), vv = make_view(v);
+  TViewMut ov = make_view_mut(out);
 
 #define DISPATCH(HB, HM)                                                      \
   hipLaunchKernelGGL((attn_fwd_kernel<HB, HM>), grid, dim3(256), 0, stream,   \
-      reinterpret_cast<const bf16_t*>(q.data_ptr()),                          \
-      reinterpret_cast<const bf16_t*>(k.data_ptr()),                          \
-      reinterpret_cast<const bf16_t*>(v.data_ptr()),                          \
-      has_bias ? reinterpret_cast<const bf16_t*>(bias->data_ptr()) : nullptr, \
+      qv, kv, vv,                                                             \
+      has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
       has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
-      reinterpret_cast<bf16_t*>(out.data_ptr()), lse.data_ptr<float>(),       \
+      ov, lse.data_ptr<float>(),                                              \
       Lq, Lk, H, (int)bias_repeat, (float)scale)
 
   if (has_bias && has_mask) DISPATCH(true, true);
@@ -751,12 +783,16 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  long bias_repeat, double scale,
                                  bool need_dbias) {
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
-  dout = dout.contiguous();
+  if (dout.stride(3) != 1) dout = dout.contiguous();
 
-  auto delta = at::empty({B, H, Lq}, q.options().dtype(at::kFloat));
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
+  auto delta = at::empty({B, H, Lq}, lse.options());
+  // gradients in (B, L, H, DH) memory, viewed (B, H, L, DH)
+  auto dq_m = at::empty({B, Lq, H, DH}, q.options());
+  auto dk_m = at::empty({B, Lk, H, DH}, q.options());
+  auto dv_m = at::empty({B, Lk, H, DH}, q.options());
+  auto dq = dq_m.permute({0, 2, 1, 3});
+  auto dk = dk_m.permute({0, 2, 1, 3});
+  auto dv = dv_m.permute({0, 2, 1, 3});
   at::Tensor dbias;
   if (need_dbias) {
     dbias = at::zeros({B / bias_repeat, H, Lq, Lk},
@@ -764,18 +800,22 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   }
   const bool has_bias = bias.has_value();
   const bool has_mask = mask.has_value();
-  at::Tensor mask_u8;
+  at::Tensor bias_c, mask_u8;
+  if (has_bias) bias_c = bias->contiguous();
   if (has_mask) mask_u8 = mask->to(at::kByte).contiguous();
 
   auto stream = at::cuda::getCurrentHIPStream();
+  TView qv = make_view(q), kvv = make_view(k), vv = make_view(v);
+  TView dov = make_view(dout), outv = make_view(out);
+  TViewMut dqv = make_view_mut(dq), dkv = make_view_mut(dk),
+           dvv = make_view_mut(dv);
+
   {
     const long rows = (long)B * H * Lq;
     const int block = 256;
     const long grid = (rows * 64 + block - 1) / block;
     hipLaunchKernelGGL(attn_delta_kernel, dim3(grid), dim3(block), 0, stream,
-                       reinterpret_cast<const bf16_t*>(dout.data_ptr()),
-                       reinterpret_cast<const bf16_t*>(out.data_ptr()),
-                       delta.data_ptr<float>(), rows);
+                       dov, outv, delta.data_ptr<float>(), H, Lq, rows);
   }
 
   dim3 grid_q((Lq + BQ - 1) / BQ, B * H);
@@ -783,15 +823,10 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
 
 #define DISPATCH_DQ(HB, HM)                                                   \
   hipLaunchKernelGGL((attn_bwd_dq_kernel<HB, HM>), grid_q, dim3(256), 0,      \
-      stream,                                                                 \
-      reinterpret_cast<const bf16_t*>(q.data_ptr()),                          \
-      reinterpret_cast<const bf16_t*>(k.data_ptr()),                          \
-      reinterpret_cast<const bf16_t*>(v.data_ptr()),                          \
-      has_bias ? reinterpret_cast<const bf16_t*>(bias->data_ptr()) : nullptr, \
+      stream, qv, kvv, vv,                                                    \
+      has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
       has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
-      reinterpret_cast<const bf16_t*>(dout.data_ptr()),                       \
-      lse.data_ptr<float>(), delta.data_ptr<float>(),                         \
-      reinterpret_cast<bf16_t*>(dq.data_ptr()),                               \
+      dov, lse.data_ptr<float>(), delta.data_ptr<float>(), dqv,               \
       Lq, Lk, H, (int)bias_repeat, (float)scale)
 
   if (has_bias && has_mask) DISPATCH_DQ(true, true);
@@ -802,16 +837,10 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
 
 #define DISPATCH_DKV(HB, HM, DB)                                              \
   hipLaunchKernelGGL((attn_bwd_dkv_kernel<HB, HM, DB>), grid_k, dim3(256), 0, \
-      stream,                                                                 \
-      reinterpret_cast<const bf16_t*>(q.data_ptr()),                          \
-      reinterpret_cast<const bf16_t*>(k.data_ptr()),                          \
-      reinterpret_cast<const bf16_t*>(v.data_ptr()),                          \
-      has_bias ? reinterpret_cast<const bf16_t*>(bias->data_ptr()) : nullptr, \
+      stream, qv, kvv, vv,                                                    \
+      has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
       has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
-      reinterpret_cast<const bf16_t*>(dout.data_ptr()),                       \
-      lse.data_ptr<float>(), delta.data_ptr<float>(),                         \
-      reinterpret_cast<bf16_t*>(dk.data_ptr()),                               \
-      reinterpret_cast<bf16_t*>(dv.data_ptr()),                               \
+      dov, lse.data_ptr<float>(), delta.data_ptr<float>(), dkv, dvv,          \
       DB ? dbias.data_ptr<float>() : nullptr,                                 \
       Lq, Lk, H, (int)bias_repeat, (float)scale)
 

@@ -11,6 +11,8 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
 at::Tensor geglu_fwd(at::Tensor x);
 at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x);
 at::Tensor dist_buckets(at::Tensor coords, at::Tensor boundaries);
+at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g);
+std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x, at::Tensor g);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
@@ -28,6 +30,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward (gfx950)");
   m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward (gfx950)");
   m.def("dist_buckets", &dist_buckets, "fused cdist+bucketize (gfx950)");
+  m.def("gatemul_fwd", &gatemul_fwd, "fused x*sigmoid(g) forward (gfx950)");
+  m.def("gatemul_bwd", &gatemul_bwd, "fused x*sigmoid(g) backward (gfx950)");
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (gfx950)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
         py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"));

@@ -59,20 +59,24 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
 }
 
 // dx = rstd * (dy*w - mean_j(dy*w) - xhat * mean_j(dy*w*xhat))
-// dw_partial / db_partial accumulated per-block then reduced on device.
-template <typename T, int VEC>
+// dw/db: each thread owns fixed columns across its block's rows — it
+// accumulates locally in registers and issues ONE atomicAdd per column
+// at the end (grid-many adds per column total, distinct-address mostly).
+template <typename T, int VEC, int MAXCHUNK>
 __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
                                      const float* __restrict__ w,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ rstd,
                                      T* __restrict__ dx,
-                                     float* __restrict__ dw_part,
-                                     float* __restrict__ db_part,
+                                     float* __restrict__ dw,
+                                     float* __restrict__ db,
                                      int rows, int D) {
   __shared__ float scratch[16];
-  float* dwp = dw_part + (long)blockIdx.x * D;
-  float* dbp = db_part + (long)blockIdx.x * D;
+  float dw_loc[MAXCHUNK * VEC];
+  float db_loc[MAXCHUNK * VEC];
+#pragma unroll
+  for (int i = 0; i < MAXCHUNK * VEC; ++i) dw_loc[i] = db_loc[i] = 0.f;
 
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* dyr = dy + (long)row * D;
@@ -81,7 +85,8 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
     const float m = mean[row], rs = rstd[row];
 
     float c1 = 0.f, c2 = 0.f;
-    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
+    int chunk = 0;
+    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC, ++chunk) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float g = to_f32(dyr[i + k]);
@@ -89,8 +94,13 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
         float gw = g * w[i + k];
         c1 += gw;
         c2 += gw * xhat;
-        dwp[i + k] += g * xhat;
-        dbp[i + k] += g;
+        if (chunk < MAXCHUNK) {
+          dw_loc[chunk * VEC + k] += g * xhat;
+          db_loc[chunk * VEC + k] += g;
+        } else {  // huge-D fallback: direct atomics
+          atomicAdd(&dw[i + k], g * xhat);
+          atomicAdd(&db[i + k], g);
+        }
       }
     }
     c1 = block_reduce_sum(c1, scratch) / D;
@@ -106,16 +116,16 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
     }
     __syncthreads();
   }
-}
 
-__global__ void reduce_partials_kernel(const float* __restrict__ part,
-                                       float* __restrict__ out,
-                                       int nblocks, int D) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= D) return;
-  float acc = 0.f;
-  for (int bidx = 0; bidx < nblocks; ++bidx) acc += part[(long)bidx * D + i];
-  out[i] = acc;
+  int chunk = 0;
+  for (int i = threadIdx.x * VEC; i < D && chunk < MAXCHUNK;
+       i += blockDim.x * VEC, ++chunk) {
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) {
+      atomicAdd(&dw[i + k], dw_loc[chunk * VEC + k]);
+      atomicAdd(&db[i + k], db_loc[chunk * VEC + k]);
+    }
+  }
 }
 
 int pick_grid(int rows) {
@@ -176,42 +186,33 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
 
   const int block = 256;
   const int grid = pick_grid(rows);
-  auto dw_part = at::zeros({grid, D}, x.options().dtype(at::kFloat));
-  auto db_part = at::zeros({grid, D}, x.options().dtype(at::kFloat));
+  auto dw = at::zeros({D}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({D}, x.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
 
+  // MAXCHUNK covers D <= 256*VEC*MAXCHUNK in registers (D<=2048 for
+  // bf16); beyond that the kernel falls back to per-row atomics.
 #define LAUNCH(T, VEC)                                                     \
-  hipLaunchKernelGGL((layernorm_bwd_kernel<T, VEC>), dim3(grid),           \
+  hipLaunchKernelGGL((layernorm_bwd_kernel<T, VEC, 1>), dim3(grid),        \
                      dim3(block), 0, stream,                               \
                      reinterpret_cast<const T*>(dy.data_ptr()),            \
                      reinterpret_cast<const T*>(x.data_ptr()),             \
                      wf.data_ptr<float>(), mean.data_ptr<float>(),         \
                      rstd.data_ptr<float>(),                               \
                      reinterpret_cast<T*>(dx.data_ptr()),                  \
-                     dw_part.data_ptr<float>(), db_part.data_ptr<float>(), \
+                     dw.data_ptr<float>(), db.data_ptr<float>(),           \
                      (int)rows, D)
 
-  const bool vec8 = (D % 8) == 0;
+  const bool vec8 = (D % 8) == 0 && D <= 2048;
   if (x.scalar_type() == at::kBFloat16) {
     if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if (vec8) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if ((D % 4) == 0 && D <= 1024) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
     if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "layernorm_bwd: unsupported dtype");
   }
 #undef LAUNCH
-
-  auto dw = at::empty({D}, x.options().dtype(at::kFloat));
-  auto db = at::empty({D}, x.options().dtype(at::kFloat));
-  const int rblock = 256;
-  const int rgrid = (D + rblock - 1) / rblock;
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(rgrid), dim3(rblock), 0,
-                     stream, dw_part.data_ptr<float>(), dw.data_ptr<float>(),
-                     grid, D);
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(rgrid), dim3(rblock), 0,
-                     stream, db_part.data_ptr<float>(), db.data_ptr<float>(),
-                     grid, D);
   return {dx, dw.to(w.scalar_type()), db.to(w.scalar_type())};
 }

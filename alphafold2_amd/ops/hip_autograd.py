@@ -55,7 +55,8 @@ class _AttentionFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, bias, mask, bias_repeat, scale):
         ext = _load_ext()
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        # q/k/v are consumed through their strides (no permute copies);
+        # only the innermost head-dim must be dense
         bias_c = bias.contiguous() if bias is not None else None
         mask_c = mask.contiguous() if mask is not None else None
         out, lse = ext.attn_fwd(q, k, v, bias_c, mask_c, bias_repeat, scale)
@@ -75,7 +76,7 @@ class _AttentionFn(torch.autograd.Function):
         q, k, v, out, lse = saved[:5]
         bias = saved[5] if ctx.has_bias else None
         need_dbias = ctx.bias_requires_grad
-        rets = ext.attn_bwd(dout.contiguous(), q, k, v, out, lse, bias,
+        rets = ext.attn_bwd(dout, q, k, v, out, lse, bias,
                             ctx.mask, ctx.bias_repeat, ctx.scale, need_dbias)
         dq, dk, dv = rets[:3]
         dbias = rets[3].to(bias.dtype) if need_dbias else None
@@ -90,6 +91,26 @@ def hip_attention_core(q, k, v, bias=None, mask=None, context_mask=None,
         key_mask = key_mask.to(torch.uint8)
     scale = q.shape[-1] ** -0.5
     return _AttentionFn.apply(q, k, v, bias, key_mask, bias_repeat, scale)
+
+
+class _GateMulFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, g):
+        ext = _load_ext()
+        x, g = x.contiguous(), g.contiguous()
+        ctx.save_for_backward(x, g)
+        return ext.gatemul_fwd(x, g)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _load_ext()
+        x, g = ctx.saved_tensors
+        dx, dg = ext.gatemul_bwd(dy.contiguous(), x, g)
+        return dx, dg
+
+
+def hip_gatemul(x, g):
+    return _GateMulFn.apply(x, g)
 
 
 def hip_outer_product_mean(left, right, mask=None, eps=1e-5):

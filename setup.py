@@ -23,6 +23,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "geglu.hip"),
         os.path.join(HIP_DIR, "distbucket.hip"),
         os.path.join(HIP_DIR, "attention.hip"),
+        os.path.join(HIP_DIR, "gatemul.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
