@@ -274,6 +274,15 @@ class Alphafold2(nn.Module):
         else:
             raise ValueError('either MSA or embedds must be given')
 
+        # under autocast, run the trunk streams natively in the compute
+        # dtype: embeddings come out fp32 and would otherwise drag every
+        # residual/LN/elementwise op through fp32 (2x HBM traffic) plus
+        # per-Linear input casts
+        if torch.is_autocast_enabled() and x.is_cuda:
+            ac = torch.get_autocast_gpu_dtype()
+            x = x.to(ac)
+            m = m.to(ac)
+
         # pairwise representation: outer sum + relative position embedding
         x_left, x_right = self.to_pairwise_repr(x).chunk(2, dim=-1)
         x = ops.pair_outer_sum(x_left, x_right)  # (b, i, j, d)

@@ -90,6 +90,38 @@ __device__ __forceinline__ void stage_tile(const bf16_t* __restrict__ g,
   }
 }
 
+// stage a [rows<=64][<=64] bf16 tile with arbitrary row stride into a
+// swizzled LDS tile; used for the pair-bias tile in the dK/dV kernel
+// (its transposed per-lane reads would otherwise scatter 2B loads).
+__device__ __forceinline__ void stage_tile_rowstride(
+    const bf16_t* __restrict__ g, long row_stride, int rows, int cols,
+    char* lds) {
+  const int tid = threadIdx.x;
+  if ((row_stride % 8) == 0 && (((uintptr_t)g) & 15) == 0) {
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      int idx = tid + pass * 256;
+      int row = idx >> 3;
+      int c16 = (idx & 7) << 4;
+      float4 val = {0, 0, 0, 0};
+      if (row < rows && c16 < cols * (int)sizeof(bf16_t)) {
+        val = *reinterpret_cast<const float4*>(
+            reinterpret_cast<const char*>(g + row * row_stride) + c16);
+      }
+      *reinterpret_cast<float4*>(lds + swz(row, c16)) = val;
+    }
+  } else {  // unaligned fallback: scalar u16 staging
+    for (int idx = tid; idx < 64 * 64; idx += 256) {
+      int row = idx >> 6;
+      int col = idx & 63;
+      bf16_t val = (bf16_t)0.f;
+      if (row < rows && col < cols) val = g[row * row_stride + col];
+      *reinterpret_cast<bf16_t*>(
+          lds + swz(row, col * (int)sizeof(bf16_t))) = val;
+    }
+  }
+}
+
 // read an 8-bf16 A/B fragment (k = (lane>>4)*8 + j) for tile row `row`,
 // k-block `kblk` (32 wide) from a swizzled LDS tile
 __device__ __forceinline__ bf16x8 frag_row(const char* lds, int row,
@@ -511,6 +543,7 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
   __shared__ char q_lds[BQ * ROWB];
   __shared__ char do_lds[BQ * ROWB];
   __shared__ char s_lds[NWAVES][16 * ROWB];
+  __shared__ char b_lds[BQ * ROWB];  // bias tile [q][kv], swizzled
   __shared__ float lse_lds[BQ];
   __shared__ float delta_lds[BQ];
   __shared__ unsigned char m_lds[BK];
@@ -575,6 +608,10 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
     __syncthreads();
     stage_tile(q_g + (long)t * BQ * q.rs, q.rs, q_rows, q_lds);
     stage_tile(do_g + (long)t * BQ * dout.rs, dout.rs, q_rows, do_lds);
+    if (HAS_BIAS) {
+      stage_tile_rowstride(bias_g + (long)t * BQ * Lk + (long)ktile * BK,
+                           Lk, q_rows, min(BK, Lk - ktile * BK), b_lds);
+    }
     if (threadIdx.x < BQ) {
       const int qq = t * BQ + threadIdx.x;
       lse_lds[threadIdx.x] = (threadIdx.x < q_rows) ? lse_g[qq] : NEG_INF;
@@ -605,8 +642,8 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
         const int kvrow = wave * 16 + (lane >> 4) * 4 + reg;
         float val = pt[c][reg] * scale;
         if (HAS_BIAS && krow_ok_c[reg] && qcol < q_rows)
-          val += to_f32(bias_g[(long)(t * BQ + qcol) * Lk
-                               + ktile * BK + kvrow]);
+          val += to_f32(*reinterpret_cast<const bf16_t*>(
+              b_lds + swz(qcol, kvrow * (int)sizeof(bf16_t))));
         pt[c][reg] = (krow_ok_c[reg] && qcol < q_rows && l > NEG_INF)
             ? __expf(val - l) : 0.f;
       }
@@ -661,19 +698,13 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
       const float dl = delta_lds[qcol];
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int kvrow = wave * 16 + (lane >> 4) * 4 + reg;
-        // dL/d(S*scale + bias) — the bias gradient has NO scale factor;
-        // scale applies only on the path back into Q K^T
+        // dL/d(S*scale + bias): stored UNscaled (it IS the bias
+        // gradient); the 1/sqrt(d) scale is applied to dK at the
+        // epilogue instead
         float ds_total = pt[c][reg] * (dpt[c][reg] - dl);
-        if (NEED_DBIAS && qcol < q_rows && kvrow < kv_rows) {
-          // fold over bias_repeat with fp32 atomics (sum over the
-          // repeat group is the correct gradient)
-          atomicAdd(&dbias_g[(long)(t * BQ + qcol) * Lk
-                             + ktile * BK + kvrow], ds_total);
-        }
         const int row = (lane >> 4) * 4 + reg;
         *reinterpret_cast<bf16_t*>(sw + swz(row, qcol * (int)sizeof(bf16_t)))
-            = (bf16_t)(ds_total * scale);
+            = (bf16_t)ds_total;
       }
     }
     __builtin_amdgcn_s_waitcnt(0);
@@ -681,6 +712,27 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
 #pragma unroll
     for (int kblk = 0; kblk < 2; ++kblk)
       dst_frag[kblk] = frag_row(sw, lane & 15, kblk);
+
+    if (NEED_DBIAS) {
+      // cooperative transposed read of the full 64x64 dS tile from the
+      // four per-wave scratches -> coalesced fp32 atomics into dbias
+      // (fold over bias_repeat); thread -> (q row, 16-wide kv chunk)
+      __syncthreads();
+      const int qq = threadIdx.x >> 2;         // 0..63
+      const int kv0 = (threadIdx.x & 3) << 4;  // 0,16,32,48
+      if (qq < q_rows) {
+        float* drow = dbias_g + (long)(t * BQ + qq) * Lk + (long)ktile * BK;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int kvr = kv0 + j;
+          if (kvr < kv_rows) {
+            float dsv = to_f32(*reinterpret_cast<const bf16_t*>(
+                s_lds[kvr >> 4] + swz(kvr & 15, qq * (int)sizeof(bf16_t))));
+            atomicAdd(&drow[kvr], dsv);
+          }
+        }
+      }
+    }
 
     // dK += dS^T Q : B col = d, k = q -> strided col reads of Q
 #pragma unroll
@@ -705,7 +757,7 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
         dk_g[(long)row * dk.rs + c * 16 + (lane & 15)] =
-            (bf16_t)dk_acc[c][reg];
+            (bf16_t)(dk_acc[c][reg] * scale);
         dv_g[(long)row * dv.rs + c * 16 + (lane & 15)] =
             (bf16_t)dv_acc[c][reg];
       }

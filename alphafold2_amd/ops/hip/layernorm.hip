@@ -3,9 +3,13 @@
 // The Evoformer pre-norms every block input (reference alphafold2.py:
 // FeedForward :84, AxialAttention :224, TriangleMultiplicative :297,
 // OuterMean :338) over tensors up to (b, n, n, D) — a pure memory-bound
-// op.  One workgroup per row, fp32 accumulation, vectorized 8-wide bf16
-// loads on the fast path (CDNA guide G13: hipcc does not auto-vectorize
-// bf16 loads).
+// op.  Parallelization: a GROUP of lanes (16/32/64, sized to D) owns one
+// row; 256-thread blocks process 256/GROUP rows concurrently, so all
+// lanes stay busy at small D (D=256 would leave 7/8 idle with a
+// block-per-row scheme).  Row statistics reduce with shfl_xor inside the
+// group — no LDS.  fp32 accumulation, vectorized 8-wide bf16 loads
+// (CDNA guide G13).  dw/db accumulate in registers per thread and issue
+// one atomicAdd per column per block at the end.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -13,56 +17,62 @@
 
 namespace {
 
-template <typename T, int VEC>
+template <int GROUP>
+__device__ __forceinline__ float group_sum(float v) {
+#pragma unroll
+  for (int off = GROUP / 2; off > 0; off >>= 1) v += __shfl_xor(v, off, GROUP);
+  return v;
+}
+
+template <typename T, int VEC, int GROUP>
 __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
                                      const float* __restrict__ w,
                                      const float* __restrict__ b,
                                      T* __restrict__ y,
                                      float* __restrict__ mean_out,
                                      float* __restrict__ rstd_out,
-                                     int rows, int D, float eps) {
-  __shared__ float scratch[16];
-  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
-    const T* xr = x + (long)row * D;
-    T* yr = y + (long)row * D;
+                                     long rows, int D, float eps) {
+  const int RPB = blockDim.x / GROUP;
+  const int lane = threadIdx.x % GROUP;
+  const int grp = threadIdx.x / GROUP;
+
+  for (long row = (long)blockIdx.x * RPB + grp; row < rows;
+       row += (long)gridDim.x * RPB) {
+    const T* xr = x + row * D;
+    T* yr = y + row * D;
 
     float sum = 0.f, sumsq = 0.f;
-    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
-      float v[VEC];
+    for (int i = lane * VEC; i < D; i += GROUP * VEC) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        v[k] = to_f32(xr[i + k]);
-        sum += v[k];
-        sumsq += v[k] * v[k];
+        float v = to_f32(xr[i + k]);
+        sum += v;
+        sumsq += v * v;
       }
     }
-    sum = block_reduce_sum(sum, scratch);
-    sumsq = block_reduce_sum(sumsq, scratch);
+    sum = group_sum<GROUP>(sum);
+    sumsq = group_sum<GROUP>(sumsq);
 
     const float mean = sum / D;
     const float var = sumsq / D - mean * mean;
     const float rstd = rsqrtf(var + eps);
-    if (threadIdx.x == 0) {
+    if (lane == 0) {
       mean_out[row] = mean;
       rstd_out[row] = rstd;
     }
 
-    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
+    for (int i = lane * VEC; i < D; i += GROUP * VEC) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float xhat = (to_f32(xr[i + k]) - mean) * rstd;
         yr[i + k] = from_f32<T>(xhat * w[i + k] + b[i + k]);
       }
     }
-    __syncthreads();
   }
 }
 
 // dx = rstd * (dy*w - mean_j(dy*w) - xhat * mean_j(dy*w*xhat))
-// dw/db: each thread owns fixed columns across its block's rows — it
-// accumulates locally in registers and issues ONE atomicAdd per column
-// at the end (grid-many adds per column total, distinct-address mostly).
-template <typename T, int VEC, int MAXCHUNK>
+template <typename T, int VEC, int GROUP, int MAXCHUNK>
 __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
                                      const float* __restrict__ w,
@@ -71,22 +81,26 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
                                      T* __restrict__ dx,
                                      float* __restrict__ dw,
                                      float* __restrict__ db,
-                                     int rows, int D) {
-  __shared__ float scratch[16];
+                                     long rows, int D) {
+  const int RPB = blockDim.x / GROUP;
+  const int lane = threadIdx.x % GROUP;
+  const int grp = threadIdx.x / GROUP;
+
   float dw_loc[MAXCHUNK * VEC];
   float db_loc[MAXCHUNK * VEC];
 #pragma unroll
   for (int i = 0; i < MAXCHUNK * VEC; ++i) dw_loc[i] = db_loc[i] = 0.f;
 
-  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
-    const T* dyr = dy + (long)row * D;
-    const T* xr = x + (long)row * D;
-    T* dxr = dx + (long)row * D;
+  for (long row = (long)blockIdx.x * RPB + grp; row < rows;
+       row += (long)gridDim.x * RPB) {
+    const T* dyr = dy + row * D;
+    const T* xr = x + row * D;
+    T* dxr = dx + row * D;
     const float m = mean[row], rs = rstd[row];
 
     float c1 = 0.f, c2 = 0.f;
     int chunk = 0;
-    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC, ++chunk) {
+    for (int i = lane * VEC; i < D; i += GROUP * VEC, ++chunk) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float g = to_f32(dyr[i + k]);
@@ -97,16 +111,16 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
         if (chunk < MAXCHUNK) {
           dw_loc[chunk * VEC + k] += g * xhat;
           db_loc[chunk * VEC + k] += g;
-        } else {  // huge-D fallback: direct atomics
+        } else {  // huge-D fallback
           atomicAdd(&dw[i + k], g * xhat);
           atomicAdd(&db[i + k], g);
         }
       }
     }
-    c1 = block_reduce_sum(c1, scratch) / D;
-    c2 = block_reduce_sum(c2, scratch) / D;
+    c1 = group_sum<GROUP>(c1) / D;
+    c2 = group_sum<GROUP>(c2) / D;
 
-    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
+    for (int i = lane * VEC; i < D; i += GROUP * VEC) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float g = to_f32(dyr[i + k]);
@@ -114,24 +128,65 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
         dxr[i + k] = from_f32<T>(rs * (g * w[i + k] - c1 - xhat * c2));
       }
     }
-    __syncthreads();
   }
 
-  int chunk = 0;
-  for (int i = threadIdx.x * VEC; i < D && chunk < MAXCHUNK;
-       i += blockDim.x * VEC, ++chunk) {
+  // one atomic per owned column per thread; groups in the block first
+  // combine across their rows via LDS to cut the atomic count by RPB
+  __shared__ float red[256 * VEC > 4096 ? 1 : 256 * VEC];  // dw then db
+  const bool lds_combine = (256 * VEC <= 4096) && (D <= GROUP * VEC);
+  if (lds_combine) {
+    // common fast path: each thread owns exactly VEC columns (chunk 0)
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      atomicAdd(&dw[i + k], dw_loc[chunk * VEC + k]);
-      atomicAdd(&db[i + k], db_loc[chunk * VEC + k]);
+    for (int k = 0; k < VEC; ++k) red[threadIdx.x * VEC + k] = dw_loc[k];
+    __syncthreads();
+    if (grp == 0) {
+      for (int g2 = 1; g2 < RPB; ++g2)
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          dw_loc[k] += red[(g2 * GROUP + lane) * VEC + k];
+      const int i = lane * VEC;
+      if (i < D)
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) atomicAdd(&dw[i + k], dw_loc[k]);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) red[threadIdx.x * VEC + k] = db_loc[k];
+    __syncthreads();
+    if (grp == 0) {
+      for (int g2 = 1; g2 < RPB; ++g2)
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          db_loc[k] += red[(g2 * GROUP + lane) * VEC + k];
+      const int i = lane * VEC;
+      if (i < D)
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) atomicAdd(&db[i + k], db_loc[k]);
+    }
+  } else {
+    int chunk = 0;
+    for (int i = lane * VEC; i < D && chunk < MAXCHUNK;
+         i += GROUP * VEC, ++chunk) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        atomicAdd(&dw[i + k], dw_loc[chunk * VEC + k]);
+        atomicAdd(&db[i + k], db_loc[chunk * VEC + k]);
+      }
     }
   }
 }
 
-int pick_grid(int rows) {
-  // memory-bound: cap the grid and grid-stride (guide §6 G11)
-  const int cap = 256 * 8;
-  return rows < cap ? rows : cap;
+int pick_group(int D, int VEC) {
+  int per_row = (D + VEC - 1) / VEC;
+  if (per_row <= 16) return 16;
+  if (per_row <= 32) return 32;
+  return 64;
+}
+
+long pick_grid(long rows, int rpb) {
+  long blocks = (rows + rpb - 1) / rpb;
+  const long cap = 256 * 8;
+  return blocks < cap ? blocks : cap;
 }
 
 }  // namespace
@@ -148,29 +203,36 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w,
   auto bf = b.to(at::kFloat).contiguous();
 
   const int block = 256;
-  const int grid = pick_grid(rows);
   auto stream = at::cuda::getCurrentHIPStream();
 
-#define LAUNCH(T, VEC)                                                     \
-  hipLaunchKernelGGL((layernorm_fwd_kernel<T, VEC>), dim3(grid),           \
-                     dim3(block), 0, stream,                               \
-                     reinterpret_cast<const T*>(x.data_ptr()),             \
+#define LAUNCH_G(T, VEC, GROUP)                                            \
+  hipLaunchKernelGGL((layernorm_fwd_kernel<T, VEC, GROUP>),                \
+                     dim3(pick_grid(rows, block / GROUP)), dim3(block), 0, \
+                     stream, reinterpret_cast<const T*>(x.data_ptr()),     \
                      wf.data_ptr<float>(), bf.data_ptr<float>(),           \
                      reinterpret_cast<T*>(y.data_ptr()),                   \
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
-                     (int)rows, D, (float)eps)
+                     rows, D, (float)eps)
+#define LAUNCH(T, VEC)                                                     \
+  do {                                                                     \
+    int g = pick_group(D, VEC);                                            \
+    if (g == 16) LAUNCH_G(T, VEC, 16);                                     \
+    else if (g == 32) LAUNCH_G(T, VEC, 32);                                \
+    else LAUNCH_G(T, VEC, 64);                                             \
+  } while (0)
 
   const bool vec8 = (D % 8) == 0;
   if (x.scalar_type() == at::kBFloat16) {
     if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if (vec8) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if ((D % 4) == 0) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
     if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "layernorm_fwd: unsupported dtype");
   }
 #undef LAUNCH
+#undef LAUNCH_G
   return {y, mean, rstd};
 }
 
@@ -183,36 +245,41 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
   const long rows = x.numel() / D;
   auto dx = at::empty_like(x);
   auto wf = w.to(at::kFloat).contiguous();
-
-  const int block = 256;
-  const int grid = pick_grid(rows);
   auto dw = at::zeros({D}, x.options().dtype(at::kFloat));
   auto db = at::zeros({D}, x.options().dtype(at::kFloat));
+
+  const int block = 256;
   auto stream = at::cuda::getCurrentHIPStream();
 
-  // MAXCHUNK covers D <= 256*VEC*MAXCHUNK in registers (D<=2048 for
-  // bf16); beyond that the kernel falls back to per-row atomics.
-#define LAUNCH(T, VEC)                                                     \
-  hipLaunchKernelGGL((layernorm_bwd_kernel<T, VEC, 1>), dim3(grid),        \
-                     dim3(block), 0, stream,                               \
-                     reinterpret_cast<const T*>(dy.data_ptr()),            \
+  // MAXCHUNK=16 covers D <= GROUP*VEC*16 in registers
+#define LAUNCH_G(T, VEC, GROUP)                                            \
+  hipLaunchKernelGGL((layernorm_bwd_kernel<T, VEC, GROUP, 16>),            \
+                     dim3(pick_grid(rows, block / GROUP)), dim3(block), 0, \
+                     stream, reinterpret_cast<const T*>(dy.data_ptr()),    \
                      reinterpret_cast<const T*>(x.data_ptr()),             \
                      wf.data_ptr<float>(), mean.data_ptr<float>(),         \
                      rstd.data_ptr<float>(),                               \
                      reinterpret_cast<T*>(dx.data_ptr()),                  \
-                     dw.data_ptr<float>(), db.data_ptr<float>(),           \
-                     (int)rows, D)
+                     dw.data_ptr<float>(), db.data_ptr<float>(), rows, D)
+#define LAUNCH(T, VEC)                                                     \
+  do {                                                                     \
+    int g = pick_group(D, VEC);                                            \
+    if (g == 16) LAUNCH_G(T, VEC, 16);                                     \
+    else if (g == 32) LAUNCH_G(T, VEC, 32);                                \
+    else LAUNCH_G(T, VEC, 64);                                             \
+  } while (0)
 
-  const bool vec8 = (D % 8) == 0 && D <= 2048;
+  const bool vec8 = (D % 8) == 0;
   if (x.scalar_type() == at::kBFloat16) {
     if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if ((D % 4) == 0 && D <= 1024) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if ((D % 4) == 0) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
     if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "layernorm_bwd: unsupported dtype");
   }
 #undef LAUNCH
+#undef LAUNCH_G
   return {dx, dw.to(w.scalar_type()), db.to(w.scalar_type())};
 }
