@@ -237,13 +237,13 @@ class TriangleMultiplicativeModule(nn.Module):
             left = left * mask
             right = right * mask
 
-        left = left * self.left_gate(x).sigmoid()
-        right = right * self.right_gate(x).sigmoid()
+        left = ops.softclamp_gate(left, self.left_gate(x))
+        right = ops.softclamp_gate(right, self.right_gate(x))
 
         out = ops.triangle_mix(left, right, self.mix)
 
         out = self.to_out_norm(out)
-        out = out * self.out_gate(x).sigmoid()
+        out = ops.softclamp_gate(out, self.out_gate(x))
         return self.to_out(out)
 
 

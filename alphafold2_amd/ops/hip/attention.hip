@@ -714,22 +714,21 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
       dst_frag[kblk] = frag_row(sw, lane & 15, kblk);
 
     if (NEED_DBIAS) {
-      // cooperative transposed read of the full 64x64 dS tile from the
-      // four per-wave scratches -> coalesced fp32 atomics into dbias
-      // (fold over bias_repeat); thread -> (q row, 16-wide kv chunk)
+      // cooperative transposed drain of the 64x64 dS tile from the four
+      // per-wave scratches into dbias.  Lanes map to CONTIGUOUS kv
+      // addresses so each store/atomic instruction coalesces; when
+      // bias_repeat == 1 this block is the sole writer of its (q, kv)
+      // range and plain stores replace atomics.
       __syncthreads();
-      const int qq = threadIdx.x >> 2;         // 0..63
-      const int kv0 = (threadIdx.x & 3) << 4;  // 0,16,32,48
-      if (qq < q_rows) {
-        float* drow = dbias_g + (long)(t * BQ + qq) * Lk + (long)ktile * BK;
-#pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const int kvr = kv0 + j;
-          if (kvr < kv_rows) {
-            float dsv = to_f32(*reinterpret_cast<const bf16_t*>(
-                s_lds[kvr >> 4] + swz(kvr & 15, qq * (int)sizeof(bf16_t))));
-            atomicAdd(&drow[kvr], dsv);
-          }
+      const int kvr = lane;            // 0..63, contiguous per wave
+      for (int j = 0; j < 16; ++j) {
+        const int qq = j * NWAVES + wave;
+        if (qq < q_rows && kvr < kv_rows) {
+          float dsv = to_f32(*reinterpret_cast<const bf16_t*>(
+              s_lds[kvr >> 4] + swz(kvr & 15, qq * (int)sizeof(bf16_t))));
+          float* addr = dbias_g + (long)(t * BQ + qq) * Lk
+              + (long)ktile * BK + kvr;
+          if (bias_repeat == 1) *addr = dsv; else atomicAdd(addr, dsv);
         }
       }
     }

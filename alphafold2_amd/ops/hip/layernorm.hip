@@ -251,9 +251,10 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
   const int block = 256;
   auto stream = at::cuda::getCurrentHIPStream();
 
-  // MAXCHUNK=16 covers D <= GROUP*VEC*16 in registers
-#define LAUNCH_G(T, VEC, GROUP)                                            \
-  hipLaunchKernelGGL((layernorm_bwd_kernel<T, VEC, GROUP, 16>),            \
+  // MAXCHUNK sized to the actual D so accumulators stay in registers
+  // (MAXCHUNK*VEC floats x2 per thread; 16 would spill)
+#define LAUNCH_GC(T, VEC, GROUP, MC)                                       \
+  hipLaunchKernelGGL((layernorm_bwd_kernel<T, VEC, GROUP, MC>),            \
                      dim3(pick_grid(rows, block / GROUP)), dim3(block), 0, \
                      stream, reinterpret_cast<const T*>(dy.data_ptr()),    \
                      reinterpret_cast<const T*>(x.data_ptr()),             \
@@ -261,6 +262,13 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
                      rstd.data_ptr<float>(),                               \
                      reinterpret_cast<T*>(dx.data_ptr()),                  \
                      dw.data_ptr<float>(), db.data_ptr<float>(), rows, D)
+#define LAUNCH_G(T, VEC, GROUP)                                            \
+  do {                                                                     \
+    int chunks = (D + GROUP * VEC - 1) / (GROUP * VEC);                    \
+    if (chunks <= 1) LAUNCH_GC(T, VEC, GROUP, 1);                          \
+    else if (chunks <= 4) LAUNCH_GC(T, VEC, GROUP, 4);                     \
+    else LAUNCH_GC(T, VEC, GROUP, 16);                                     \
+  } while (0)
 #define LAUNCH(T, VEC)                                                     \
   do {                                                                     \
     int g = pick_group(D, VEC);                                            \
@@ -281,5 +289,6 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
   }
 #undef LAUNCH
 #undef LAUNCH_G
+#undef LAUNCH_GC
   return {dx, dw.to(w.scalar_type()), db.to(w.scalar_type())};
 }
