@@ -274,15 +274,6 @@ class Alphafold2(nn.Module):
         else:
             raise ValueError('either MSA or embedds must be given')
 
-        # under autocast, run the trunk streams natively in the compute
-        # dtype: embeddings come out fp32 and would otherwise drag every
-        # residual/LN/elementwise op through fp32 (2x HBM traffic) plus
-        # per-Linear input casts
-        if torch.is_autocast_enabled() and x.is_cuda:
-            ac = torch.get_autocast_gpu_dtype()
-            x = x.to(ac)
-            m = m.to(ac)
-
         # pairwise representation: outer sum + relative position embedding
         x_left, x_right = self.to_pairwise_repr(x).chunk(2, dim=-1)
         x = ops.pair_outer_sum(x_left, x_right)  # (b, i, j, d)
@@ -354,6 +345,15 @@ class Alphafold2(nn.Module):
             t_angle_feats = self.template_angle_mlp(templates_angles)
             m = torch.cat((m, t_angle_feats), dim=1)
             msa_mask = torch.cat((msa_mask, templates_mask), dim=1)
+
+        # under autocast, run the trunk streams natively in the compute
+        # dtype: embedding/positional additions above are fp32 and would
+        # otherwise drag every residual/LN/elementwise op in the trunk
+        # through fp32 (2x HBM traffic) plus per-Linear input casts
+        if torch.is_autocast_enabled() and x.is_cuda:
+            ac = torch.get_autocast_gpu_dtype()
+            x = x.to(ac)
+            m = m.to(ac)
 
         # extra MSAs run through their own evoformer with tied-query
         # column attention (fix over ref :790 which embeds `msa` here)

@@ -122,6 +122,32 @@ __device__ __forceinline__ void stage_tile_rowstride(
   }
 }
 
+// stage a [rows<=64][64] bf16 tile TRANSPOSED into LDS ([col][row],
+// swizzled): the PV B-operand wants V^T rows so its fragments become
+// single 16-byte ds_reads instead of 8 scalar strided reads.
+__device__ __forceinline__ void stage_tile_t(const bf16_t* __restrict__ g,
+                                             long row_stride, int rows,
+                                             char* lds) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    int idx = tid + pass * 256;
+    int row = idx >> 3;                // source row (kv)
+    int c8 = (idx & 7) << 3;           // first of 8 source cols (dv)
+    float4 val = {0, 0, 0, 0};
+    if (row < rows) {
+      val = *reinterpret_cast<const float4*>(
+          reinterpret_cast<const char*>(g + row * row_stride) + c8 * 2);
+    }
+    const bf16_t* vv = reinterpret_cast<const bf16_t*>(&val);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      *reinterpret_cast<bf16_t*>(
+          lds + swz(c8 + j, row * (int)sizeof(bf16_t))) = vv[j];
+    }
+  }
+}
+
 // read an 8-bf16 A/B fragment (k = (lane>>4)*8 + j) for tile row `row`,
 // k-block `kblk` (32 wide) from a swizzled LDS tile
 __device__ __forceinline__ bf16x8 frag_row(const char* lds, int row,
@@ -176,7 +202,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
                      float scale) {
   __shared__ char q_lds[BQ * ROWB];
   __shared__ char k_lds[BK * ROWB];
-  __shared__ char v_lds[BK * ROWB];
+  __shared__ char vt_lds[BK * ROWB];  // V transposed: [dv][kv]
   __shared__ char p_lds[NWAVES][16 * ROWB];
   __shared__ unsigned char m_lds[BK];
 
@@ -222,7 +248,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
     const int kv_rows = min(BK, Lk - t * BK);
     __syncthreads();
     stage_tile(k_g + (long)t * BK * k.rs, k.rs, kv_rows, k_lds);
-    stage_tile(v_g + (long)t * BK * v.rs, v.rs, kv_rows, v_lds);
+    stage_tile_t(v_g + (long)t * BK * v.rs, v.rs, kv_rows, vt_lds);
     if (HAS_MASK && threadIdx.x < BK) {
       m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
           ? mask[(long)batch * Lk + t * BK + threadIdx.x] : 0;
@@ -299,13 +325,13 @@ void attn_fwd_kernel(TView q, TView k, TView v,
     for (int kblk = 0; kblk < 2; ++kblk)
       p_frag[kblk] = frag_row(pw, lane & 15, kblk);
 
-    // O += P V
+    // O += P V  (B-operand = V^T rows: one b128 read per fragment)
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       f32x4 acc = o_acc[c];
 #pragma unroll
       for (int kblk = 0; kblk < 2; ++kblk) {
-        bf16x8 vf = frag_col(v_lds, c * 16 + (lane & 15), kblk);
+        bf16x8 vf = frag_row(vt_lds, c * 16 + (lane & 15), kblk);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[kblk], vf, acc,
                                                       0, 0, 0);
       }
