@@ -86,10 +86,14 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
   const int lane = threadIdx.x % GROUP;
   const int grp = threadIdx.x / GROUP;
 
-  float dw_loc[MAXCHUNK * VEC];
-  float db_loc[MAXCHUNK * VEC];
+  // STATIC indexing only — a runtime-indexed local array is allocated
+  // in scratch memory (guide common-mistake #20: 5x slowdowns)
+  float dw_loc[MAXCHUNK][VEC];
+  float db_loc[MAXCHUNK][VEC];
 #pragma unroll
-  for (int i = 0; i < MAXCHUNK * VEC; ++i) dw_loc[i] = db_loc[i] = 0.f;
+  for (int c = 0; c < MAXCHUNK; ++c)
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) dw_loc[c][k] = db_loc[c][k] = 0.f;
 
   for (long row = (long)blockIdx.x * RPB + grp; row < rows;
        row += (long)gridDim.x * RPB) {
@@ -99,8 +103,24 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
     const float m = mean[row], rs = rstd[row];
 
     float c1 = 0.f, c2 = 0.f;
-    int chunk = 0;
-    for (int i = lane * VEC; i < D; i += GROUP * VEC, ++chunk) {
+#pragma unroll
+    for (int chunk = 0; chunk < MAXCHUNK; ++chunk) {
+      const int i = (chunk * GROUP + lane) * VEC;
+      if (i < D) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          float g = to_f32(dyr[i + k]);
+          float xhat = (to_f32(xr[i + k]) - m) * rs;
+          float gw = g * w[i + k];
+          c1 += gw;
+          c2 += gw * xhat;
+          dw_loc[chunk][k] += g * xhat;
+          db_loc[chunk][k] += g;
+        }
+      }
+    }
+    // huge-D tail beyond the register-held chunks: direct atomics
+    for (int i = (MAXCHUNK * GROUP + lane) * VEC; i < D; i += GROUP * VEC) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float g = to_f32(dyr[i + k]);
@@ -108,13 +128,8 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
         float gw = g * w[i + k];
         c1 += gw;
         c2 += gw * xhat;
-        if (chunk < MAXCHUNK) {
-          dw_loc[chunk * VEC + k] += g * xhat;
-          db_loc[chunk * VEC + k] += g;
-        } else {  // huge-D fallback
-          atomicAdd(&dw[i + k], g * xhat);
-          atomicAdd(&db[i + k], g);
-        }
+        atomicAdd(&dw[i + k], g * xhat);
+        atomicAdd(&db[i + k], g);
       }
     }
     c1 = group_sum<GROUP>(c1) / D;
@@ -137,40 +152,42 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
   if (lds_combine) {
     // common fast path: each thread owns exactly VEC columns (chunk 0)
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) red[threadIdx.x * VEC + k] = dw_loc[k];
+    for (int k = 0; k < VEC; ++k) red[threadIdx.x * VEC + k] = dw_loc[0][k];
     __syncthreads();
     if (grp == 0) {
       for (int g2 = 1; g2 < RPB; ++g2)
 #pragma unroll
         for (int k = 0; k < VEC; ++k)
-          dw_loc[k] += red[(g2 * GROUP + lane) * VEC + k];
+          dw_loc[0][k] += red[(g2 * GROUP + lane) * VEC + k];
       const int i = lane * VEC;
       if (i < D)
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) atomicAdd(&dw[i + k], dw_loc[k]);
+        for (int k = 0; k < VEC; ++k) atomicAdd(&dw[i + k], dw_loc[0][k]);
     }
     __syncthreads();
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) red[threadIdx.x * VEC + k] = db_loc[k];
+    for (int k = 0; k < VEC; ++k) red[threadIdx.x * VEC + k] = db_loc[0][k];
     __syncthreads();
     if (grp == 0) {
       for (int g2 = 1; g2 < RPB; ++g2)
 #pragma unroll
         for (int k = 0; k < VEC; ++k)
-          db_loc[k] += red[(g2 * GROUP + lane) * VEC + k];
+          db_loc[0][k] += red[(g2 * GROUP + lane) * VEC + k];
       const int i = lane * VEC;
       if (i < D)
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) atomicAdd(&db[i + k], db_loc[k]);
+        for (int k = 0; k < VEC; ++k) atomicAdd(&db[i + k], db_loc[0][k]);
     }
   } else {
-    int chunk = 0;
-    for (int i = lane * VEC; i < D && chunk < MAXCHUNK;
-         i += GROUP * VEC, ++chunk) {
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) {
-        atomicAdd(&dw[i + k], dw_loc[chunk * VEC + k]);
-        atomicAdd(&db[i + k], db_loc[chunk * VEC + k]);
+    for (int chunk = 0; chunk < MAXCHUNK; ++chunk) {
+      const int i = (chunk * GROUP + lane) * VEC;
+      if (i < D) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          atomicAdd(&dw[i + k], dw_loc[chunk][k]);
+          atomicAdd(&db[i + k], db_loc[chunk][k]);
+        }
       }
     }
   }
