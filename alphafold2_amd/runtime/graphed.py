@@ -44,17 +44,20 @@ class GraphedTrainStep:
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
 
+        self.capture_error = None
         try:
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
                 self.static_loss = step_fn()
             self.graph = g
             self.graphed = True
-        except Exception:
+        except Exception as e:
             if not fallback:
                 raise
+            self.capture_error = e
             self.graph = None
             self.graphed = False
+            torch.cuda.synchronize()
 
     def __call__(self):
         if self.graphed:

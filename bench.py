@@ -122,8 +122,9 @@ def main():
         from alphafold2_amd.runtime import GraphedTrainStep
         graphed = GraphedTrainStep(step, warmup=max(2, args.warmup))
         if rank == 0:
-            print(f"# hipGraph capture: {graphed.graphed}",
-                  flush=True)
+            print(f"# hipGraph capture: {graphed.graphed}"
+                  + (f" ({graphed.capture_error})" if not graphed.graphed
+                     else ""), flush=True)
         step = graphed
 
     for _ in range(args.warmup):
