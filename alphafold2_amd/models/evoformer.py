@@ -351,13 +351,18 @@ class Evoformer(nn.Module):
     """Stack of EvoformerBlocks with per-block activation checkpointing
     as the default memory strategy during training (the reference's
     approach — alphafold2.py:466); `checkpoint_blocks=False` disables it
-    (e.g. when the reversible trunk handles memory instead)."""
+    (e.g. when the reversible trunk handles memory instead).
+
+    RNG state is only stashed per checkpoint when dropout is active —
+    the stash is a host sync that would break hipGraph step capture."""
 
     def __init__(self, *, depth, checkpoint_blocks=True, **kwargs):
         super().__init__()
         self.layers = nn.ModuleList(
             [EvoformerBlock(**kwargs) for _ in range(depth)])
         self.checkpoint_blocks = checkpoint_blocks
+        self.preserve_rng_state = (kwargs.get('attn_dropout', 0.) > 0
+                                   or kwargs.get('ff_dropout', 0.) > 0)
 
     def forward(self, x, m, mask=None, msa_mask=None):
         inp = (x, m, mask, msa_mask)
@@ -365,7 +370,8 @@ class Evoformer(nn.Module):
             and torch.is_grad_enabled()
         for layer in self.layers:
             if use_ckpt:
-                inp = checkpoint(layer, inp, use_reentrant=False)
+                inp = checkpoint(layer, inp, use_reentrant=False,
+                                 preserve_rng_state=self.preserve_rng_state)
             else:
                 inp = layer(inp)
         x, m, *_ = inp
