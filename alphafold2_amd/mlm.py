@@ -77,6 +77,11 @@ class MLM(nn.Module):
 
     def forward(self, seq_embed, original_seq, mask):
         logits = self.to_logits(seq_embed)
-        seq_logits = logits[mask]
-        seq_labels = original_seq[mask]
-        return F.cross_entropy(seq_logits, seq_labels, reduction='mean')
+        # static-shape masked mean (no boolean gather: a gather would
+        # force a device sync and break hipGraph step capture)
+        V = logits.shape[-1]
+        per_tok = F.cross_entropy(
+            logits.reshape(-1, V).float(), original_seq.reshape(-1),
+            reduction='none')
+        fmask = mask.reshape(-1).to(per_tok.dtype)
+        return (per_tok * fmask).sum() / fmask.sum().clamp(min=1)
