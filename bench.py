@@ -38,6 +38,10 @@ def parse_args():
                    help='use the reversible trunk execution mode')
     p.add_argument('--no-graph', action='store_true',
                    help='disable hipGraph step capture')
+    p.add_argument('--checkpoint-blocks', action='store_true',
+                   help='enable per-block activation checkpointing '
+                        '(needed only when activations exceed HBM; '
+                        'costs a full forward recompute)')
     p.add_argument('--device', type=str, default=None)
     return p.parse_args()
 
@@ -71,6 +75,7 @@ def main():
         dim_head=args.dim_head,
         max_seq_len=max(2048, args.crop_len),
         reversible=args.reversible,
+        checkpoint_blocks=args.checkpoint_blocks,
     ).to(device)
     model.train()
 
