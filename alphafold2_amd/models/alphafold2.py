@@ -76,9 +76,11 @@ class Alphafold2(nn.Module):
         predict_angles=False,
         symmetrize_omega=False,
         predict_coords=False,
+        structure_module_type='ipa',
         structure_module_depth=4,
         structure_module_heads=1,
         structure_module_dim_head=4,
+        structure_module_refinement_iters=None,  # alias for depth
         disable_token_embed=False,
         mlm_mask_prob=0.15,
         mlm_random_replace_token_prob=0.1,
@@ -182,23 +184,38 @@ class Alphafold2(nn.Module):
             nn.Linear(dim, constants.DISTOGRAM_BUCKETS),
         )
 
-        # structure module
+        # structure module: 'ipa' (default, reference-layout compatible)
+        # or native equivariant alternatives 'egnn' / 'se3'
+        assert structure_module_type in ('ipa', 'egnn', 'se3'), \
+            "structure_module_type must be 'ipa', 'egnn' or 'se3'"
+        if structure_module_refinement_iters is not None:
+            structure_module_depth = structure_module_refinement_iters
         self.predict_coords = predict_coords
+        self.structure_module_type = structure_module_type
         self.structure_module_depth = structure_module_depth
 
         self.msa_to_single_repr_dim = nn.Linear(dim, dim)
         self.trunk_to_pairwise_repr_dim = nn.Linear(dim, dim)
 
-        with torch_default_dtype(torch.float32):
-            self.ipa_block = IPABlock(
-                dim=dim,
-                heads=structure_module_heads,
-            )
-            self.to_quaternion_update = nn.Linear(dim, 6)
+        if structure_module_type == 'ipa':
+            with torch_default_dtype(torch.float32):
+                self.ipa_block = IPABlock(
+                    dim=dim,
+                    heads=structure_module_heads,
+                )
+                self.to_quaternion_update = nn.Linear(dim, 6)
 
-        init_zero_(self.ipa_block.attn.to_out)
+            init_zero_(self.ipa_block.attn.to_out)
 
-        self.to_points = nn.Linear(dim, 3)
+            self.to_points = nn.Linear(dim, 3)
+        else:
+            from .equivariant import EquivariantStructureModule
+            with torch_default_dtype(torch.float32):
+                self.structure_module = EquivariantStructureModule(
+                    dim, depth=structure_module_depth,
+                    kind=structure_module_type,
+                    heads=max(structure_module_heads, 4),
+                    dim_head=max(structure_module_dim_head, 16))
 
         # per-residue confidence head
         self.lddt_linear = nn.Linear(dim, 1)
@@ -397,6 +414,25 @@ class Alphafold2(nn.Module):
         original_dtype = single_repr.dtype
         single_repr = single_repr.float()
         pairwise_repr = pairwise_repr.float()
+
+        if self.structure_module_type != 'ipa':
+            # equivariant refinement (EGNN / SE3-style), fp32
+            with torch_default_dtype(torch.float32):
+                start = recyclables.coords.float() \
+                    if exists(recyclables) else None
+                single_repr, coords = self.structure_module(
+                    single_repr, pairwise_repr, mask=mask, coords=start)
+            coords = coords.type(original_dtype)
+
+            if return_recyclables:
+                rec = map(torch.detach,
+                          (coords, single_msa_repr_row, pairwise_repr))
+                ret.recyclables = Recyclables(*rec)
+            if return_aux_logits:
+                return coords, ret
+            if return_confidence:
+                return coords, self.lddt_linear(single_repr.float())
+            return coords
 
         # iterative IPA refinement in fp32 (equivariance)
         with torch_default_dtype(torch.float32):

@@ -214,3 +214,60 @@ def test_reversible_trunk():
     loss.backward()
     grads = [p.grad for p in model.net.parameters() if p.grad is not None]
     assert len(grads) > 0
+
+
+def test_egnn_structure_module():
+    model = tiny_model(predict_coords=True, structure_module_type='egnn',
+                       structure_module_depth=2)
+    seq = torch.randint(0, 21, (2, 12))
+    mask = torch.ones_like(seq).bool()
+    msa = torch.randint(0, 21, (2, 3, 12))
+    msa_mask = torch.ones_like(msa).bool()
+    coords = model(seq, msa, mask=mask, msa_mask=msa_mask)
+    assert coords.shape == (2, 12, 3)
+    coords.sum().backward()
+
+
+def test_se3_structure_module_equivariance():
+    """Rotating/translating the recycled input coords must rotate the
+    output coords identically (SE(3) equivariance of the refiner)."""
+    from alphafold2_amd.models.equivariant import EquivariantStructureModule
+    torch.manual_seed(0)
+    mod = EquivariantStructureModule(dim=16, depth=2, kind='se3').double()
+    h = torch.randn(1, 10, 16, dtype=torch.float64)
+    pair = torch.randn(1, 10, 10, 16, dtype=torch.float64)
+    x0 = torch.randn(1, 10, 3, dtype=torch.float64)
+
+    theta = torch.tensor(0.6, dtype=torch.float64)
+    R = torch.tensor([[torch.cos(theta), -torch.sin(theta), 0],
+                      [torch.sin(theta), torch.cos(theta), 0],
+                      [0., 0., 1.]], dtype=torch.float64)
+    t = torch.tensor([1., -2., 3.], dtype=torch.float64)
+
+    _, out1 = mod(h, pair, coords=x0)
+    _, out2 = mod(h, pair, coords=x0 @ R.T + t)
+    assert torch.allclose(out2, out1 @ R.T + t, atol=1e-8), \
+        (out2 - (out1 @ R.T + t)).abs().max()
+
+
+def test_egnn_equivariance():
+    from alphafold2_amd.models.equivariant import EquivariantStructureModule
+    torch.manual_seed(0)
+    mod = EquivariantStructureModule(dim=16, depth=2, kind='egnn').double()
+    h = torch.randn(1, 8, 16, dtype=torch.float64)
+    pair = torch.randn(1, 8, 8, 16, dtype=torch.float64)
+    x0 = torch.randn(1, 8, 3, dtype=torch.float64)
+    theta = torch.tensor(1.1, dtype=torch.float64)
+    R = torch.tensor([[1., 0., 0.],
+                      [0., torch.cos(theta), -torch.sin(theta)],
+                      [0., torch.sin(theta), torch.cos(theta)]],
+                     dtype=torch.float64)
+    t = torch.tensor([-1., 0.5, 2.], dtype=torch.float64)
+    _, out1 = mod(h, pair, coords=x0)
+    _, out2 = mod(h, pair, coords=x0 @ R.T + t)
+    assert torch.allclose(out2, out1 @ R.T + t, atol=1e-8)
+
+
+def test_refinement_iters_alias():
+    m = tiny_model(predict_coords=True, structure_module_refinement_iters=2)
+    assert m.structure_module_depth == 2
