@@ -368,7 +368,7 @@ class Alphafold2(nn.Module):
         # otherwise drag every residual/LN/elementwise op in the trunk
         # through fp32 (2x HBM traffic) plus per-Linear input casts
         if torch.is_autocast_enabled() and x.is_cuda:
-            ac = torch.get_autocast_gpu_dtype()
+            ac = torch.get_autocast_dtype('cuda')
             x = x.to(ac)
             m = m.to(ac)
 

@@ -148,6 +148,11 @@ class _ReversibleFunction(Function):
         ctx.blocks = blocks
         ctx.mask = mask
         ctx.msa_mask = msa_mask
+        # the reconstruction passes in backward must recompute under the
+        # same autocast regime as this forward, or dtypes diverge
+        ctx.amp_enabled = torch.is_autocast_enabled()
+        ctx.amp_dtype = torch.get_autocast_dtype('cuda') \
+            if ctx.amp_enabled else None
         with torch.no_grad():
             for block in blocks:
                 x1, x2, m1, m2 = block(
@@ -158,11 +163,15 @@ class _ReversibleFunction(Function):
 
     @staticmethod
     def backward(ctx, dx1, dx2, dm1, dm2):
+        import contextlib
+        amp = torch.autocast('cuda', dtype=ctx.amp_dtype) \
+            if ctx.amp_enabled else contextlib.nullcontext()
         y = ctx.y
         dy = (dx1, dx2, dm1, dm2)
-        for block in reversed(ctx.blocks):
-            y, dy = block.backward_pass(y, dy, mask=ctx.mask,
-                                        msa_mask=ctx.msa_mask)
+        with amp:
+            for block in reversed(ctx.blocks):
+                y, dy = block.backward_pass(y, dy, mask=ctx.mask,
+                                            msa_mask=ctx.msa_mask)
         return (*dy, None, None, None)
 
 
