@@ -255,8 +255,10 @@ void attn_fwd_kernel(TView q, TView k, TView v,
     }
     __syncthreads();
 
-    // S = Q K^T  (16 q x 64 kv per wave)
+    // S = Q K^T  (16 q x 64 kv per wave); setprio favors the MFMA
+    // cluster when co-resident waves are staging (guide T5)
     f32x4 s[4];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       f32x4 acc = {0, 0, 0, 0};
@@ -268,6 +270,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
       }
       s[c] = acc;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // scale + bias + masking (C layout: row=(lane>>4)*4+reg, col=lane&15)
 #pragma unroll

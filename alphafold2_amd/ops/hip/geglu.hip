@@ -12,21 +12,21 @@
 
 namespace {
 
+// row-outer loop: no per-element division (a runtime `idx / H` in the
+// hot loop serializes on the integer-div unit)
 template <typename T, int VEC>
 __global__ void geglu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                  long rows, int H) {
-  const long total = rows * (long)H;
-  const long stride = (long)gridDim.x * blockDim.x * VEC;
-  for (long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
-       base < total; base += stride) {
-    const long row = base / H;
-    const int col = base - row * H;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + row * (2L * H);
+    T* yr = y + row * (long)H;
+    for (int col = threadIdx.x * VEC; col < H; col += blockDim.x * VEC) {
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      float a = to_f32(xr[col + k]);
-      float g = to_f32(xr[H + col + k]);
-      y[base + k] = from_f32<T>(a * gelu_f(g));
+      for (int k = 0; k < VEC; ++k) {
+        float a = to_f32(xr[col + k]);
+        float g = to_f32(xr[H + col + k]);
+        yr[col + k] = from_f32<T>(a * gelu_f(g));
+      }
     }
   }
 }
@@ -35,21 +35,19 @@ template <typename T, int VEC>
 __global__ void geglu_bwd_kernel(const T* __restrict__ dy,
                                  const T* __restrict__ x,
                                  T* __restrict__ dx, long rows, int H) {
-  const long total = rows * (long)H;
-  const long stride = (long)gridDim.x * blockDim.x * VEC;
-  for (long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
-       base < total; base += stride) {
-    const long row = base / H;
-    const int col = base - row * H;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* dyr = dy + row * (long)H;
     const T* xr = x + row * (2L * H);
     T* dxr = dx + row * (2L * H);
+    for (int col = threadIdx.x * VEC; col < H; col += blockDim.x * VEC) {
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      float a = to_f32(xr[col + k]);
-      float g = to_f32(xr[H + col + k]);
-      float go = to_f32(dy[base + k]);
-      dxr[col + k] = from_f32<T>(go * gelu_f(g));
-      dxr[H + col + k] = from_f32<T>(go * a * gelu_grad_f(g));
+      for (int k = 0; k < VEC; ++k) {
+        float a = to_f32(xr[col + k]);
+        float g = to_f32(xr[H + col + k]);
+        float go = to_f32(dyr[col + k]);
+        dxr[col + k] = from_f32<T>(go * gelu_f(g));
+        dxr[H + col + k] = from_f32<T>(go * a * gelu_grad_f(g));
+      }
     }
   }
 }
@@ -73,8 +71,7 @@ at::Tensor geglu_fwd(at::Tensor x) {
 
 #define LAUNCH(T, VEC)                                                   \
   do {                                                                   \
-    long grid = (total / VEC + block - 1) / block;                       \
-    if (grid > 2048) grid = 2048;                                        \
+    long grid = rows < 4096 ? rows : 4096;                               \
     if (grid < 1) grid = 1;                                              \
     hipLaunchKernelGGL((geglu_fwd_kernel<T, VEC>), dim3(grid),           \
                        dim3(block), 0, stream,                           \
@@ -82,13 +79,19 @@ at::Tensor geglu_fwd(at::Tensor x) {
                        reinterpret_cast<T*>(y.data_ptr()), rows, H);     \
   } while (0)
 
-  const bool vec8 = (H % 8) == 0;
+  // adaptive VEC: keep all 256 lanes busy when H < 2048
+  const bool vec8 = (H % 8) == 0 && H >= 2048;
+  const bool vec4 = (H % 4) == 0;
   if (x.scalar_type() == at::kBFloat16) {
-    if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
+    if (vec8) LAUNCH(__hip_bfloat16, 8);
+    else if (vec4) LAUNCH(__hip_bfloat16, 4);
+    else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if (vec8) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if (vec4) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
-    if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
+    if (vec8) LAUNCH(__half, 8);
+    else if (vec4) LAUNCH(__half, 4);
+    else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "geglu_fwd: unsupported dtype");
   }
@@ -110,8 +113,7 @@ at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x) {
 
 #define LAUNCH(T, VEC)                                                   \
   do {                                                                   \
-    long grid = (total / VEC + block - 1) / block;                       \
-    if (grid > 2048) grid = 2048;                                        \
+    long grid = rows < 4096 ? rows : 4096;                               \
     if (grid < 1) grid = 1;                                              \
     hipLaunchKernelGGL((geglu_bwd_kernel<T, VEC>), dim3(grid),           \
                        dim3(block), 0, stream,                           \
@@ -120,13 +122,19 @@ at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x) {
                        reinterpret_cast<T*>(dx.data_ptr()), rows, H);    \
   } while (0)
 
-  const bool vec8 = (H % 8) == 0;
+  // adaptive VEC: keep all 256 lanes busy when H < 2048
+  const bool vec8 = (H % 8) == 0 && H >= 2048;
+  const bool vec4 = (H % 4) == 0;
   if (x.scalar_type() == at::kBFloat16) {
-    if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
+    if (vec8) LAUNCH(__hip_bfloat16, 8);
+    else if (vec4) LAUNCH(__hip_bfloat16, 4);
+    else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if (vec8) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if (vec4) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
-    if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
+    if (vec8) LAUNCH(__half, 8);
+    else if (vec4) LAUNCH(__half, 4);
+    else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "geglu_bwd: unsupported dtype");
   }
