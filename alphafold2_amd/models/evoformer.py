@@ -143,7 +143,7 @@ class Attention(nn.Module):
         # sigmoid output gating (init to identity)
         if gates is None:
             gates = self.gating(x)
-        out = ops.softclamp_gate(out.contiguous(), gates.contiguous())
+        out = ops.softclamp_gate(out, gates)
         return self.to_out(out)
 
 
@@ -261,13 +261,13 @@ class TriangleMultiplicativeModule(nn.Module):
             left = left * mask
             right = right * mask
 
-        left = ops.softclamp_gate(left.contiguous(), lg.contiguous())
-        right = ops.softclamp_gate(right.contiguous(), rg.contiguous())
+        left = ops.softclamp_gate(left, lg)
+        right = ops.softclamp_gate(right, rg)
 
         out = ops.triangle_mix(left, right, self.mix)
 
         out = self.to_out_norm(out)
-        out = ops.softclamp_gate(out, og.contiguous())
+        out = ops.softclamp_gate(out, og)
         return self.to_out(out)
 
 

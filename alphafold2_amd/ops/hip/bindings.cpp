@@ -11,8 +11,9 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
 at::Tensor geglu_fwd(at::Tensor x);
 at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x);
 at::Tensor dist_buckets(at::Tensor coords, at::Tensor boundaries);
-at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g);
-std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x, at::Tensor g);
+at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs);
+std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x, at::Tensor g,
+                                    long xs, long gs);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,

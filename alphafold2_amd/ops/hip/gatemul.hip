@@ -2,8 +2,9 @@
 //
 // Used by attention output gating (reference alphafold2.py:184-185) and
 // the three triangle-multiplicative gates (:306-311).  One kernel
-// instead of sigmoid + mul (and their backward chains); vectorized
-// 8-wide bf16 loads (guide G13).
+// instead of sigmoid + mul; inputs may be row-strided SLICES of a fused
+// projection (no contiguous() copies); row-group parallelism keeps all
+// lanes on 16-byte accesses at any channel width.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -15,111 +16,147 @@ __device__ __forceinline__ float sigmoid_f(float x) {
   return 1.0f / (1.0f + __expf(-x));
 }
 
-template <typename T, int VEC>
-__global__ void gatemul_fwd_kernel(const T* __restrict__ x,
-                                   const T* __restrict__ g,
-                                   T* __restrict__ y, long total) {
-  const long stride = (long)gridDim.x * blockDim.x * VEC;
-  for (long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
-       base < total; base += stride) {
+template <typename T, int VEC, int GROUP>
+__global__ void gatemul_fwd_kernel(const T* __restrict__ x, long xs,
+                                   const T* __restrict__ g, long gs,
+                                   T* __restrict__ y,
+                                   long rows, int C) {
+  const int RPB = blockDim.x / GROUP;
+  const int lane = threadIdx.x % GROUP;
+  const int grp = threadIdx.x / GROUP;
+  for (long row = (long)blockIdx.x * RPB + grp; row < rows;
+       row += (long)gridDim.x * RPB) {
+    const T* xr = x + row * xs;
+    const T* gr = g + row * gs;
+    T* yr = y + row * (long)C;
+    for (int i = lane * VEC; i < C; i += GROUP * VEC) {
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      y[base + k] = from_f32<T>(to_f32(x[base + k]) *
-                                sigmoid_f(to_f32(g[base + k])));
+      for (int k = 0; k < VEC; ++k) {
+        yr[i + k] = from_f32<T>(to_f32(xr[i + k]) *
+                                sigmoid_f(to_f32(gr[i + k])));
+      }
     }
   }
 }
 
-template <typename T, int VEC>
+template <typename T, int VEC, int GROUP>
 __global__ void gatemul_bwd_kernel(const T* __restrict__ dy,
-                                   const T* __restrict__ x,
-                                   const T* __restrict__ g,
+                                   const T* __restrict__ x, long xs,
+                                   const T* __restrict__ g, long gs,
                                    T* __restrict__ dx, T* __restrict__ dg,
-                                   long total) {
-  const long stride = (long)gridDim.x * blockDim.x * VEC;
-  for (long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
-       base < total; base += stride) {
+                                   long rows, int C) {
+  const int RPB = blockDim.x / GROUP;
+  const int lane = threadIdx.x % GROUP;
+  const int grp = threadIdx.x / GROUP;
+  for (long row = (long)blockIdx.x * RPB + grp; row < rows;
+       row += (long)gridDim.x * RPB) {
+    const T* dyr = dy + row * (long)C;
+    const T* xr = x + row * xs;
+    const T* gr = g + row * gs;
+    T* dxr = dx + row * (long)C;
+    T* dgr = dg + row * (long)C;
+    for (int i = lane * VEC; i < C; i += GROUP * VEC) {
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      float go = to_f32(dy[base + k]);
-      float xv = to_f32(x[base + k]);
-      float s = sigmoid_f(to_f32(g[base + k]));
-      dx[base + k] = from_f32<T>(go * s);
-      dg[base + k] = from_f32<T>(go * xv * s * (1.f - s));
+      for (int k = 0; k < VEC; ++k) {
+        float go = to_f32(dyr[i + k]);
+        float xv = to_f32(xr[i + k]);
+        float s = sigmoid_f(to_f32(gr[i + k]));
+        dxr[i + k] = from_f32<T>(go * s);
+        dgr[i + k] = from_f32<T>(go * xv * s * (1.f - s));
+      }
     }
   }
+}
+
+int gm_group(int C, int VEC) {
+  int per_row = (C + VEC - 1) / VEC;
+  if (per_row <= 16) return 16;
+  if (per_row <= 32) return 32;
+  return 64;
+}
+
+long gm_grid(long rows, int rpb) {
+  long blocks = (rows + rpb - 1) / rpb;
+  const long cap = 4096;
+  return blocks < cap ? blocks : (cap > 0 ? cap : 1);
 }
 
 }  // namespace
 
-at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g) {
-  TORCH_CHECK(x.is_contiguous() && g.is_contiguous(),
-              "gatemul_fwd: inputs must be contiguous");
-  TORCH_CHECK(x.sizes() == g.sizes(), "gatemul_fwd: shape mismatch");
-  auto y = at::empty_like(x);
-  const long total = x.numel();
+// x/g: (rows, C) with uniform row strides xs/gs (elements); y contiguous
+at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs) {
+  const int C = x.size(-1);
+  const long rows = x.numel() / C;
+  auto y = at::empty(x.sizes(), x.options());
   const int block = 256;
   auto stream = at::cuda::getCurrentHIPStream();
 
-#define LAUNCH(T, VEC)                                                   \
-  do {                                                                   \
-    long grid = (total / VEC + block - 1) / block;                       \
-    if (grid > 2048) grid = 2048;                                        \
-    if (grid < 1) grid = 1;                                              \
-    hipLaunchKernelGGL((gatemul_fwd_kernel<T, VEC>), dim3(grid),         \
-                       dim3(block), 0, stream,                           \
-                       reinterpret_cast<const T*>(x.data_ptr()),         \
-                       reinterpret_cast<const T*>(g.data_ptr()),         \
-                       reinterpret_cast<T*>(y.data_ptr()), total);       \
+#define LAUNCH_G(T, VEC, GROUP)                                             \
+  hipLaunchKernelGGL((gatemul_fwd_kernel<T, VEC, GROUP>),                   \
+                     dim3(gm_grid(rows, block / GROUP)), dim3(block), 0,    \
+                     stream, reinterpret_cast<const T*>(x.data_ptr()), xs,  \
+                     reinterpret_cast<const T*>(g.data_ptr()), gs,          \
+                     reinterpret_cast<T*>(y.data_ptr()), rows, C)
+#define LAUNCH(T, VEC)                                                      \
+  do {                                                                      \
+    int gg = gm_group(C, VEC);                                              \
+    if (gg == 16) LAUNCH_G(T, VEC, 16);                                     \
+    else if (gg == 32) LAUNCH_G(T, VEC, 32);                                \
+    else LAUNCH_G(T, VEC, 64);                                              \
   } while (0)
 
-  const bool vec8 = (total % 8) == 0;
   if (x.scalar_type() == at::kBFloat16) {
-    if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
+    if ((C % 8) == 0) LAUNCH(__hip_bfloat16, 8);
+    else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if ((total % 4) == 0) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if ((C % 4) == 0) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
-    if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
+    if ((C % 8) == 0) LAUNCH(__half, 8); else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "gatemul_fwd: unsupported dtype");
   }
 #undef LAUNCH
+#undef LAUNCH_G
   return y;
 }
 
 std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x,
-                                    at::Tensor g) {
-  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous() && g.is_contiguous());
-  auto dx = at::empty_like(x);
-  auto dg = at::empty_like(g);
-  const long total = x.numel();
+                                    at::Tensor g, long xs, long gs) {
+  TORCH_CHECK(dy.is_contiguous(), "gatemul_bwd: dy must be contiguous");
+  const int C = x.size(-1);
+  const long rows = x.numel() / C;
+  auto dx = at::empty(x.sizes(), x.options());
+  auto dg = at::empty(x.sizes(), x.options());
   const int block = 256;
   auto stream = at::cuda::getCurrentHIPStream();
 
-#define LAUNCH(T, VEC)                                                   \
-  do {                                                                   \
-    long grid = (total / VEC + block - 1) / block;                       \
-    if (grid > 2048) grid = 2048;                                        \
-    if (grid < 1) grid = 1;                                              \
-    hipLaunchKernelGGL((gatemul_bwd_kernel<T, VEC>), dim3(grid),         \
-                       dim3(block), 0, stream,                           \
-                       reinterpret_cast<const T*>(dy.data_ptr()),        \
-                       reinterpret_cast<const T*>(x.data_ptr()),         \
-                       reinterpret_cast<const T*>(g.data_ptr()),         \
-                       reinterpret_cast<T*>(dx.data_ptr()),              \
-                       reinterpret_cast<T*>(dg.data_ptr()), total);      \
+#define LAUNCH_G(T, VEC, GROUP)                                             \
+  hipLaunchKernelGGL((gatemul_bwd_kernel<T, VEC, GROUP>),                   \
+                     dim3(gm_grid(rows, block / GROUP)), dim3(block), 0,    \
+                     stream, reinterpret_cast<const T*>(dy.data_ptr()),     \
+                     reinterpret_cast<const T*>(x.data_ptr()), xs,          \
+                     reinterpret_cast<const T*>(g.data_ptr()), gs,          \
+                     reinterpret_cast<T*>(dx.data_ptr()),                   \
+                     reinterpret_cast<T*>(dg.data_ptr()), rows, C)
+#define LAUNCH(T, VEC)                                                      \
+  do {                                                                      \
+    int gg = gm_group(C, VEC);                                              \
+    if (gg == 16) LAUNCH_G(T, VEC, 16);                                     \
+    else if (gg == 32) LAUNCH_G(T, VEC, 32);                                \
+    else LAUNCH_G(T, VEC, 64);                                              \
   } while (0)
 
-  const bool vec8 = (total % 8) == 0;
   if (x.scalar_type() == at::kBFloat16) {
-    if (vec8) LAUNCH(__hip_bfloat16, 8); else LAUNCH(__hip_bfloat16, 1);
+    if ((C % 8) == 0) LAUNCH(__hip_bfloat16, 8);
+    else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if ((total % 4) == 0) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if ((C % 4) == 0) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
-    if (vec8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
+    if ((C % 8) == 0) LAUNCH(__half, 8); else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "gatemul_bwd: unsupported dtype");
   }
 #undef LAUNCH
+#undef LAUNCH_G
   return {dx, dg};
 }

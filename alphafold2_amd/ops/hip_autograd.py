@@ -93,19 +93,46 @@ def hip_attention_core(q, k, v, bias=None, mask=None, context_mask=None,
     return _AttentionFn.apply(q, k, v, bias, key_mask, bias_repeat, scale)
 
 
+def _uniform_row_stride(t):
+    """Row stride (elements) if the tensor enumerates as (rows, C) with
+    one uniform stride (e.g. a channel slice of a fused projection);
+    None if the layout is irregular."""
+    C = t.shape[-1]
+    if t.stride(-1) != 1:
+        return None
+    if t.dim() == 1:
+        return C
+    rs = t.stride(-2)
+    mult = t.shape[-2]
+    for i in range(t.dim() - 3, -1, -1):
+        if t.shape[i] != 1 and t.stride(i) != rs * mult:
+            return None
+        mult *= t.shape[i]
+    return rs
+
+
 class _GateMulFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, g):
         ext = _load_ext()
-        x, g = x.contiguous(), g.contiguous()
+        xs = _uniform_row_stride(x)
+        gs = _uniform_row_stride(g)
+        if xs is None:
+            x = x.contiguous()
+            xs = x.shape[-1]
+        if gs is None:
+            g = g.contiguous()
+            gs = g.shape[-1]
         ctx.save_for_backward(x, g)
-        return ext.gatemul_fwd(x, g)
+        ctx.strides = (xs, gs)
+        return ext.gatemul_fwd(x, g, xs, gs)
 
     @staticmethod
     def backward(ctx, dy):
         ext = _load_ext()
         x, g = ctx.saved_tensors
-        dx, dg = ext.gatemul_bwd(dy.contiguous(), x, g)
+        xs, gs = ctx.strides
+        dx, dg = ext.gatemul_bwd(dy.contiguous(), x, g, xs, gs)
         return dx, dg
 
 
