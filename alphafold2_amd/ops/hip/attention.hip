@@ -398,6 +398,7 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
   __shared__ char q_lds[BQ * ROWB];
   __shared__ char do_lds[BQ * ROWB];
   __shared__ char k_lds[BK * ROWB];
+  __shared__ char kt_lds[BK * ROWB];  // K transposed: [d][kv]
   __shared__ char v_lds[BK * ROWB];
   __shared__ char s_lds[NWAVES][16 * ROWB];
   __shared__ unsigned char m_lds[BK];
@@ -453,6 +454,7 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
     const int kv_rows = min(BK, Lk - t * BK);
     __syncthreads();
     stage_tile(k_g + (long)t * BK * k.rs, k.rs, kv_rows, k_lds);
+    stage_tile_t(k_g + (long)t * BK * k.rs, k.rs, kv_rows, kt_lds);
     stage_tile(v_g + (long)t * BK * v.rs, v.rs, kv_rows, v_lds);
     if (HAS_MASK && threadIdx.x < BK) {
       m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
@@ -521,18 +523,20 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
     for (int kblk = 0; kblk < 2; ++kblk)
       ds_frag[kblk] = frag_row(sw, lane & 15, kblk);
 
-    // dQ += dS K : B col = d, k = kv -> strided col reads of K
+    // dQ += dS K : B col = d, k = kv -> b128 rows of the K^T tile
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       f32x4 acc = dq_acc[c];
 #pragma unroll
       for (int kblk = 0; kblk < 2; ++kblk) {
-        bf16x8 kf = frag_col(k_lds, c * 16 + (lane & 15), kblk);
+        bf16x8 kf = frag_row(kt_lds, c * 16 + (lane & 15), kblk);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag[kblk], kf, acc,
                                                       0, 0, 0);
       }
       dq_acc[c] = acc;
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   bf16_t* dq_g = dq.base(batch, head) + (long)qtile * BQ * dq.rs;
@@ -570,7 +574,9 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
   __shared__ char k_lds[BK * ROWB];
   __shared__ char v_lds[BK * ROWB];
   __shared__ char q_lds[BQ * ROWB];
+  __shared__ char qt_lds[BQ * ROWB];   // Q transposed: [d][q]
   __shared__ char do_lds[BQ * ROWB];
+  __shared__ char dot_lds[BQ * ROWB];  // dO transposed: [dv][q]
   __shared__ char s_lds[NWAVES][16 * ROWB];
   __shared__ char b_lds[BQ * ROWB];  // bias tile [q][kv], swizzled
   __shared__ float lse_lds[BQ];
@@ -636,7 +642,9 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
     const int q_rows = min(BQ, Lq - t * BQ);
     __syncthreads();
     stage_tile(q_g + (long)t * BQ * q.rs, q.rs, q_rows, q_lds);
+    stage_tile_t(q_g + (long)t * BQ * q.rs, q.rs, q_rows, qt_lds);
     stage_tile(do_g + (long)t * BQ * dout.rs, dout.rs, q_rows, do_lds);
+    stage_tile_t(do_g + (long)t * BQ * dout.rs, dout.rs, q_rows, dot_lds);
     if (HAS_BIAS) {
       stage_tile_rowstride(bias_g + (long)t * BQ * Lk + (long)ktile * BK,
                            Lk, q_rows, min(BK, Lk - ktile * BK), b_lds);
@@ -695,17 +703,19 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
 #pragma unroll
     for (int kblk = 0; kblk < 2; ++kblk)
       pt_frag[kblk] = frag_row(sw, lane & 15, kblk);
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       f32x4 acc = dv_acc[c];
 #pragma unroll
       for (int kblk = 0; kblk < 2; ++kblk) {
-        bf16x8 dof = frag_col(do_lds, c * 16 + (lane & 15), kblk);
+        bf16x8 dof = frag_row(dot_lds, c * 16 + (lane & 15), kblk);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag[kblk], dof, acc,
                                                       0, 0, 0);
       }
       dv_acc[c] = acc;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // dP^T = V dO^T : A = V (k = dv), B col = q, k = dv (contiguous)
 #pragma unroll
@@ -762,18 +772,20 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
       }
     }
 
-    // dK += dS^T Q : B col = d, k = q -> strided col reads of Q
+    // dK += dS^T Q : B col = d, k = q -> b128 rows of the Q^T tile
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       f32x4 acc = dk_acc[c];
 #pragma unroll
       for (int kblk = 0; kblk < 2; ++kblk) {
-        bf16x8 qf = frag_col(q_lds, c * 16 + (lane & 15), kblk);
+        bf16x8 qf = frag_row(qt_lds, c * 16 + (lane & 15), kblk);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dst_frag[kblk], qf, acc,
                                                       0, 0, 0);
       }
       dk_acc[c] = acc;
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   bf16_t* dk_g = dk.base(batch, head) + (long)ktile * BK * dk.rs;
