@@ -1,2 +1,3 @@
 from .graphed import GraphedTrainStep
 from .checkpointing import save_checkpoint, load_checkpoint
+from .profiling import StepTimer, kernel_stats_summary, profile_trace

@@ -146,3 +146,15 @@ def test_graphed_step_cpu_fallback_unavailable():
         pytest.skip('GPU present')
     with pytest.raises(AssertionError):
         GraphedTrainStep(lambda: torch.zeros(1))
+
+
+def test_kernel_stats_summary():
+    from alphafold2_amd.runtime import kernel_stats_summary
+    import glob
+    csvs = glob.glob('profiles/*kernel_stats*.csv')
+    if not csvs:
+        import pytest
+        pytest.skip('no committed profiles')
+    rows = kernel_stats_summary(csvs[0], top=5)
+    assert len(rows) == 5
+    assert rows[0]['pct'] >= rows[1]['pct']
