@@ -119,6 +119,20 @@ class Attention(nn.Module):
                 self.gating.bias])
             inner = self.to_q.weight.shape[0]
             fused = F.linear(x, w, bias_cat)
+
+            if tie_dim is None:
+                # packed fast path: the fused kernel consumes q/k/v as
+                # strided slices and writes one packed grad (no
+                # split-backward concatenation)
+                out = ops.attention_core_packed(
+                    fused, h, inner, bias=attn_bias, mask=mask,
+                    bias_repeat=attn_bias_repeat)
+                if out is not None:
+                    out = out.transpose(-2, -3).reshape(*x.shape[:-1], -1)
+                    gates = fused.narrow(-1, 3 * inner, inner)
+                    out = ops.softclamp_gate(out, gates)
+                    return self.to_out(out)
+
             q, k, v, gates = fused.split([inner, inner, inner, inner], dim=-1)
         else:
             q = self.to_q(x)

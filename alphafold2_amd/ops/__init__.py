@@ -22,6 +22,6 @@ AF2AMD_ALLOW_EAGER_GPU=1 to override.
 """
 from .dispatch import (  # noqa: F401
     hip_ops_available, using_hip,
-    attention_core, geglu, outer_product_mean, triangle_mix,
+    attention_core, attention_core_packed, geglu, outer_product_mean, triangle_mix,
     pair_outer_sum, distance_buckets, layer_norm, softclamp_gate,
 )

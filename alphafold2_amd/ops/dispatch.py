@@ -86,6 +86,26 @@ def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
                                 context_mask=context_mask, tie_dim=tie_dim)
 
 
+def attention_core_packed(packed, heads, inner, bias=None, mask=None,
+                          context_mask=None, bias_repeat=1):
+    """Self-attention consuming a packed [q|k|v|...] projection (B, L, W).
+    HIP-only fast path; returns None if not fusable (caller falls back
+    to the split path)."""
+    key_mask = context_mask if context_mask is not None else mask
+    fusable = (
+        packed.dtype == torch.bfloat16
+        and (inner // heads) == 64
+        and (bias is None or bias.dtype == torch.bfloat16)
+        and packed.is_cuda
+        and using_hip(packed, 'attn_fwd')
+    )
+    if not fusable:
+        return None
+    from .hip_autograd import hip_attention_packed
+    return hip_attention_packed(packed, heads, inner, bias=bias,
+                                mask=key_mask, bias_repeat=bias_repeat)
+
+
 def geglu(x):
     if using_hip(x, 'geglu_fwd'):
         from .hip_autograd import hip_geglu

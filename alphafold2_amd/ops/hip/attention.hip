@@ -873,18 +873,25 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
                                  long bias_repeat, double scale,
-                                 bool need_dbias) {
+                                 bool need_dbias,
+                                 c10::optional<at::Tensor> dq_out,
+                                 c10::optional<at::Tensor> dk_out,
+                                 c10::optional<at::Tensor> dv_out) {
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
   if (dout.stride(3) != 1) dout = dout.contiguous();
 
   auto delta = at::empty({B, H, Lq}, lse.options());
-  // gradients in (B, L, H, DH) memory, viewed (B, H, L, DH)
-  auto dq_m = at::empty({B, Lq, H, DH}, q.options());
-  auto dk_m = at::empty({B, Lk, H, DH}, q.options());
-  auto dv_m = at::empty({B, Lk, H, DH}, q.options());
-  auto dq = dq_m.permute({0, 2, 1, 3});
-  auto dk = dk_m.permute({0, 2, 1, 3});
-  auto dv = dv_m.permute({0, 2, 1, 3});
+  // gradients written through strided views — either caller-provided
+  // slices of a packed buffer (no concatenation in autograd) or fresh
+  // (B, L, H, DH) memory viewed (B, H, L, DH)
+  at::Tensor dq, dk, dv;
+  if (dq_out.has_value()) {
+    dq = *dq_out; dk = *dk_out; dv = *dv_out;
+  } else {
+    dq = at::empty({B, Lq, H, DH}, q.options()).permute({0, 2, 1, 3});
+    dk = at::empty({B, Lk, H, DH}, q.options()).permute({0, 2, 1, 3});
+    dv = at::empty({B, Lk, H, DH}, q.options()).permute({0, 2, 1, 3});
+  }
   at::Tensor dbias;
   if (need_dbias) {
     dbias = at::zeros({B / bias_repeat, H, Lq, Lk},

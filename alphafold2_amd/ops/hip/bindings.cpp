@@ -23,7 +23,10 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
                                  long bias_repeat, double scale,
-                                 bool need_dbias);
+                                 bool need_dbias,
+                                 c10::optional<at::Tensor> dq_out,
+                                 c10::optional<at::Tensor> dk_out,
+                                 c10::optional<at::Tensor> dv_out);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm forward (gfx950)");
@@ -36,5 +39,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (gfx950)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
         py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"));
-  m.def("attn_bwd", &attn_bwd, "fused flash attention backward (gfx950)");
+  m.def("attn_bwd", &attn_bwd, "fused flash attention backward (gfx950)",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("out"), py::arg("lse"), py::arg("bias"), py::arg("mask"),
+        py::arg("bias_repeat"), py::arg("scale"), py::arg("need_dbias"),
+        py::arg("dq_out") = c10::nullopt, py::arg("dk_out") = c10::nullopt,
+        py::arg("dv_out") = c10::nullopt);
 }

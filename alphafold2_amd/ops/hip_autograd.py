@@ -93,6 +93,68 @@ def hip_attention_core(q, k, v, bias=None, mask=None, context_mask=None,
     return _AttentionFn.apply(q, k, v, bias, key_mask, bias_repeat, scale)
 
 
+class _AttentionPackedFn(torch.autograd.Function):
+    """Self-attention over a PACKED projection tensor (B, L, W) holding
+    [q | k | v | ...] channel blocks of h*dh each.  The backward writes
+    dq/dk/dv directly into slices of ONE grad buffer, so autograd never
+    concatenates three per-slice gradients (the split-backward cat was
+    ~5% of a training step)."""
+
+    @staticmethod
+    def forward(ctx, packed, heads, inner, bias, mask, bias_repeat, scale):
+        ext = _load_ext()
+        B, L = packed.shape[0], packed.shape[1]
+        dh = inner // heads
+
+        def view(off):
+            return packed.narrow(-1, off, inner)                 .view(B, L, heads, dh).permute(0, 2, 1, 3)
+
+        q, k, v = view(0), view(inner), view(2 * inner)
+        bias_c = bias.contiguous() if bias is not None else None
+        mask_c = mask.contiguous() if mask is not None else None
+        out, lse = ext.attn_fwd(q, k, v, bias_c, mask_c, bias_repeat, scale)
+        ctx.save_for_backward(packed, out, lse,
+                              *([bias_c] if bias_c is not None else []))
+        ctx.has_bias = bias_c is not None
+        ctx.mask = mask_c
+        ctx.meta = (heads, inner, bias_repeat, scale)
+        ctx.bias_requires_grad = bias is not None and bias.requires_grad
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _load_ext()
+        saved = ctx.saved_tensors
+        packed, out, lse = saved[:3]
+        bias = saved[3] if ctx.has_bias else None
+        heads, inner, bias_repeat, scale = ctx.meta
+        B, L, W = packed.shape
+        dh = inner // heads
+
+        dpacked = torch.zeros_like(packed)
+
+        def view(t, off):
+            return t.narrow(-1, off, inner)                 .view(B, L, heads, dh).permute(0, 2, 1, 3)
+
+        q, k, v = view(packed, 0), view(packed, inner), view(packed, 2 * inner)
+        dq, dk, dv = (view(dpacked, 0), view(dpacked, inner),
+                      view(dpacked, 2 * inner))
+        need_dbias = ctx.bias_requires_grad
+        rets = ext.attn_bwd(dout, q, k, v, out, lse, bias, ctx.mask,
+                            bias_repeat, scale, need_dbias,
+                            dq_out=dq, dk_out=dk, dv_out=dv)
+        dbias = rets[3].to(bias.dtype) if need_dbias else None
+        return dpacked, None, None, dbias, None, None, None
+
+
+def hip_attention_packed(packed, heads, inner, bias=None, mask=None,
+                         bias_repeat=1):
+    key_mask = mask.to(torch.uint8) if mask is not None else None
+    scale = (inner // heads) ** -0.5
+    return _AttentionPackedFn.apply(packed, heads, inner, bias, key_mask,
+                                    bias_repeat, scale)
+
+
 def _uniform_row_stride(t):
     """Row stride (elements) if the tensor enumerates as (rows, C) with
     one uniform stride (e.g. a channel slice of a fused projection);
