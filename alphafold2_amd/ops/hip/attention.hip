@@ -368,17 +368,24 @@ void attn_fwd_kernel(TView q, TView k, TView v,
 __global__ void attn_delta_kernel(TView dout, TView out,
                                   float* __restrict__ delta,
                                   int heads, int Lq, long rows) {
-  // one 64-lane wave per (b, h, l) row of DH=64
-  const long row = (long)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
+  // 8 rows per 64-lane wave: lane l covers row (l>>3), 8 channels
+  // starting at (l&7)*8 — 16-byte loads, 1 KiB per wave transaction
+  const long row = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 3;
   if (row >= rows) return;
-  const int lane = threadIdx.x & 63;
+  const int sub = threadIdx.x & 7;
   const int l = row % Lq;
-  const int bh = row / Lq;
-  const int b = bh / heads, h = bh - (long)(bh / heads) * heads;
-  float v = to_f32(dout.base(b, h)[(long)l * dout.rs + lane]) *
-            to_f32(out.base(b, h)[(long)l * out.rs + lane]);
-  v = wave_reduce_sum(v);
-  if (lane == 0) delta[row] = v;
+  const long bh = row / Lq;
+  const int b = bh / heads, h = bh - (bh / heads) * heads;
+  const bf16_t* dp = dout.base(b, h) + (long)l * dout.rs + sub * 8;
+  const bf16_t* op = out.base(b, h) + (long)l * out.rs + sub * 8;
+  bf16x8 dv8 = *reinterpret_cast<const bf16x8*>(dp);
+  bf16x8 ov8 = *reinterpret_cast<const bf16x8*>(op);
+  float v = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) v += to_f32(dv8[j]) * to_f32(ov8[j]);
+#pragma unroll
+  for (int off = 4; off > 0; off >>= 1) v += __shfl_xor(v, off, 8);
+  if (sub == 0) delta[row] = v;
 }
 
 // ---------------------------------------------------------------------------
@@ -912,7 +919,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   {
     const long rows = (long)B * H * Lq;
     const int block = 256;
-    const long grid = (rows * 64 + block - 1) / block;
+    const long grid = (rows * 8 + block - 1) / block;
     hipLaunchKernelGGL(attn_delta_kernel, dim3(grid), dim3(block), 0, stream,
                        dov, outv, delta.data_ptr<float>(), H, Lq, rows);
   }
