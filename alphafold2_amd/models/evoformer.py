@@ -255,8 +255,6 @@ class TriangleMultiplicativeModule(nn.Module):
 
     def forward(self, x, mask=None):
         assert x.shape[1] == x.shape[2], 'feature map must be symmetrical'
-        if exists(mask):
-            mask = mask[..., None]
 
         x = self.norm(x)
 
@@ -271,12 +269,10 @@ class TriangleMultiplicativeModule(nn.Module):
         fused = F.linear(x, w, bias_cat)
         left, right, lg, rg, og = fused.split([hdim] * 5, dim=-1)
 
-        if exists(mask):
-            left = left * mask
-            right = right * mask
-
-        left = ops.softclamp_gate(left, lg)
-        right = ops.softclamp_gate(right, rg)
+        # the pair mask folds into the fused gate kernel (no separate
+        # full-tensor mask-multiply passes)
+        left = ops.softclamp_gate(left, lg, row_mask=mask)
+        right = ops.softclamp_gate(right, rg, row_mask=mask)
 
         out = ops.triangle_mix(left, right, self.mix)
 

@@ -145,8 +145,13 @@ def layer_norm(x, weight, bias, eps=1e-5):
     return eager.layer_norm(x, weight, bias, eps)
 
 
-def softclamp_gate(x, gates):
+def softclamp_gate(x, gates, row_mask=None):
+    """out = x * sigmoid(gates) (* row_mask broadcast per row).
+    row_mask: optional bool with shape == x.shape[:-1]."""
     if x.shape == gates.shape and using_hip(x, 'gatemul_fwd'):
         from .hip_autograd import hip_gatemul
-        return hip_gatemul(x, gates)
-    return eager.softclamp_gate(x, gates)
+        return hip_gatemul(x, gates, row_mask)
+    out = eager.softclamp_gate(x, gates)
+    if row_mask is not None:
+        out = out * row_mask.unsqueeze(-1).to(out.dtype)
+    return out

@@ -11,9 +11,11 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
 at::Tensor geglu_fwd(at::Tensor x);
 at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x);
 at::Tensor dist_buckets(at::Tensor coords, at::Tensor boundaries);
-at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs);
+at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs,
+                       c10::optional<at::Tensor> rowmask);
 std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x, at::Tensor g,
-                                    long xs, long gs);
+                                    long xs, long gs,
+                                    c10::optional<at::Tensor> rowmask);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
@@ -34,8 +36,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward (gfx950)");
   m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward (gfx950)");
   m.def("dist_buckets", &dist_buckets, "fused cdist+bucketize (gfx950)");
-  m.def("gatemul_fwd", &gatemul_fwd, "fused x*sigmoid(g) forward (gfx950)");
-  m.def("gatemul_bwd", &gatemul_bwd, "fused x*sigmoid(g) backward (gfx950)");
+  m.def("gatemul_fwd", &gatemul_fwd, "fused x*sigmoid(g) forward (gfx950)",
+        py::arg("x"), py::arg("g"), py::arg("xs"), py::arg("gs"),
+        py::arg("rowmask") = c10::nullopt);
+  m.def("gatemul_bwd", &gatemul_bwd, "fused x*sigmoid(g) backward (gfx950)",
+        py::arg("dy"), py::arg("x"), py::arg("g"), py::arg("xs"),
+        py::arg("gs"), py::arg("rowmask") = c10::nullopt);
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (gfx950)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
         py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"));

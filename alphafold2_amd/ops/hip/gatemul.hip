@@ -16,9 +16,12 @@ __device__ __forceinline__ float sigmoid_f(float x) {
   return 1.0f / (1.0f + __expf(-x));
 }
 
-template <typename T, int VEC, int GROUP>
+// rowmask: optional per-row multiplier (0/1 byte) fused in — saves the
+// separate mask-multiply passes in the triangle module
+template <typename T, int VEC, int GROUP, bool RM>
 __global__ void gatemul_fwd_kernel(const T* __restrict__ x, long xs,
                                    const T* __restrict__ g, long gs,
+                                   const unsigned char* __restrict__ rowmask,
                                    T* __restrict__ y,
                                    long rows, int C) {
   const int RPB = blockDim.x / GROUP;
@@ -29,20 +32,22 @@ __global__ void gatemul_fwd_kernel(const T* __restrict__ x, long xs,
     const T* xr = x + row * xs;
     const T* gr = g + row * gs;
     T* yr = y + row * (long)C;
+    const float mv = RM ? (float)rowmask[row] : 1.f;
     for (int i = lane * VEC; i < C; i += GROUP * VEC) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        yr[i + k] = from_f32<T>(to_f32(xr[i + k]) *
+        yr[i + k] = from_f32<T>(to_f32(xr[i + k]) * mv *
                                 sigmoid_f(to_f32(gr[i + k])));
       }
     }
   }
 }
 
-template <typename T, int VEC, int GROUP>
+template <typename T, int VEC, int GROUP, bool RM>
 __global__ void gatemul_bwd_kernel(const T* __restrict__ dy,
                                    const T* __restrict__ x, long xs,
                                    const T* __restrict__ g, long gs,
+                                   const unsigned char* __restrict__ rowmask,
                                    T* __restrict__ dx, T* __restrict__ dg,
                                    long rows, int C) {
   const int RPB = blockDim.x / GROUP;
@@ -55,10 +60,11 @@ __global__ void gatemul_bwd_kernel(const T* __restrict__ dy,
     const T* gr = g + row * gs;
     T* dxr = dx + row * (long)C;
     T* dgr = dg + row * (long)C;
+    const float mv = RM ? (float)rowmask[row] : 1.f;
     for (int i = lane * VEC; i < C; i += GROUP * VEC) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float go = to_f32(dyr[i + k]);
+        float go = to_f32(dyr[i + k]) * mv;
         float xv = to_f32(xr[i + k]);
         float s = sigmoid_f(to_f32(gr[i + k]));
         dxr[i + k] = from_f32<T>(go * s);
@@ -84,19 +90,28 @@ long gm_grid(long rows, int rpb) {
 }  // namespace
 
 // x/g: (rows, C) with uniform row strides xs/gs (elements); y contiguous
-at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs) {
+at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs,
+                       c10::optional<at::Tensor> rowmask) {
   const int C = x.size(-1);
   const long rows = x.numel() / C;
   auto y = at::empty(x.sizes(), x.options());
   const int block = 256;
   auto stream = at::cuda::getCurrentHIPStream();
+  const bool rm = rowmask.has_value();
+  const unsigned char* rm_ptr =
+      rm ? rowmask->data_ptr<unsigned char>() : nullptr;
 
-#define LAUNCH_G(T, VEC, GROUP)                                             \
-  hipLaunchKernelGGL((gatemul_fwd_kernel<T, VEC, GROUP>),                   \
+#define LAUNCH_G2(T, VEC, GROUP, RM)                                        \
+  hipLaunchKernelGGL((gatemul_fwd_kernel<T, VEC, GROUP, RM>),               \
                      dim3(gm_grid(rows, block / GROUP)), dim3(block), 0,    \
                      stream, reinterpret_cast<const T*>(x.data_ptr()), xs,  \
-                     reinterpret_cast<const T*>(g.data_ptr()), gs,          \
+                     reinterpret_cast<const T*>(g.data_ptr()), gs, rm_ptr,  \
                      reinterpret_cast<T*>(y.data_ptr()), rows, C)
+#define LAUNCH_G(T, VEC, GROUP)                                             \
+  do {                                                                      \
+    if (rm) LAUNCH_G2(T, VEC, GROUP, true);                                 \
+    else LAUNCH_G2(T, VEC, GROUP, false);                                   \
+  } while (0)
 #define LAUNCH(T, VEC)                                                      \
   do {                                                                      \
     int gg = gm_group(C, VEC);                                              \
@@ -117,11 +132,13 @@ at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs) {
   }
 #undef LAUNCH
 #undef LAUNCH_G
+#undef LAUNCH_G2
   return y;
 }
 
 std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x,
-                                    at::Tensor g, long xs, long gs) {
+                                    at::Tensor g, long xs, long gs,
+                                    c10::optional<at::Tensor> rowmask) {
   TORCH_CHECK(dy.is_contiguous(), "gatemul_bwd: dy must be contiguous");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
@@ -129,15 +146,23 @@ std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x,
   auto dg = at::empty(x.sizes(), x.options());
   const int block = 256;
   auto stream = at::cuda::getCurrentHIPStream();
+  const bool rm = rowmask.has_value();
+  const unsigned char* rm_ptr =
+      rm ? rowmask->data_ptr<unsigned char>() : nullptr;
 
-#define LAUNCH_G(T, VEC, GROUP)                                             \
-  hipLaunchKernelGGL((gatemul_bwd_kernel<T, VEC, GROUP>),                   \
+#define LAUNCH_G2(T, VEC, GROUP, RM)                                        \
+  hipLaunchKernelGGL((gatemul_bwd_kernel<T, VEC, GROUP, RM>),               \
                      dim3(gm_grid(rows, block / GROUP)), dim3(block), 0,    \
                      stream, reinterpret_cast<const T*>(dy.data_ptr()),     \
                      reinterpret_cast<const T*>(x.data_ptr()), xs,          \
-                     reinterpret_cast<const T*>(g.data_ptr()), gs,          \
+                     reinterpret_cast<const T*>(g.data_ptr()), gs, rm_ptr,  \
                      reinterpret_cast<T*>(dx.data_ptr()),                   \
                      reinterpret_cast<T*>(dg.data_ptr()), rows, C)
+#define LAUNCH_G(T, VEC, GROUP)                                             \
+  do {                                                                      \
+    if (rm) LAUNCH_G2(T, VEC, GROUP, true);                                 \
+    else LAUNCH_G2(T, VEC, GROUP, false);                                   \
+  } while (0)
 #define LAUNCH(T, VEC)                                                      \
   do {                                                                      \
     int gg = gm_group(C, VEC);                                              \
@@ -158,5 +183,6 @@ std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x,
   }
 #undef LAUNCH
 #undef LAUNCH_G
+#undef LAUNCH_G2
   return {dx, dg};
 }

@@ -175,7 +175,7 @@ def _uniform_row_stride(t):
 
 class _GateMulFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, g):
+    def forward(ctx, x, g, rowmask):
         ext = _load_ext()
         xs = _uniform_row_stride(x)
         gs = _uniform_row_stride(g)
@@ -185,21 +185,24 @@ class _GateMulFn(torch.autograd.Function):
         if gs is None:
             g = g.contiguous()
             gs = g.shape[-1]
+        rm = rowmask.reshape(-1).to(torch.uint8).contiguous() \
+            if rowmask is not None else None
         ctx.save_for_backward(x, g)
         ctx.strides = (xs, gs)
-        return ext.gatemul_fwd(x, g, xs, gs)
+        ctx.rm = rm
+        return ext.gatemul_fwd(x, g, xs, gs, rm)
 
     @staticmethod
     def backward(ctx, dy):
         ext = _load_ext()
         x, g = ctx.saved_tensors
         xs, gs = ctx.strides
-        dx, dg = ext.gatemul_bwd(dy.contiguous(), x, g, xs, gs)
-        return dx, dg
+        dx, dg = ext.gatemul_bwd(dy.contiguous(), x, g, xs, gs, ctx.rm)
+        return dx, dg, None
 
 
-def hip_gatemul(x, g):
-    return _GateMulFn.apply(x, g)
+def hip_gatemul(x, g, rowmask=None):
+    return _GateMulFn.apply(x, g, rowmask)
 
 
 def hip_outer_product_mean(left, right, mask=None, eps=1e-5):
