@@ -52,12 +52,14 @@ def relax_structure(coords, mask=None, iters=50, lr=0.05,
             bm = (mask[:, 1:] & mask[:, :-1]).float()
             bond_loss = bond_loss * bm
         d = torch.cdist(x, x)
-        eye = torch.eye(d.shape[-1], device=d.device, dtype=torch.bool)
+        n = d.shape[-1]
+        # exclude self and bonded neighbours from the clash term
+        excl = torch.eye(n, device=d.device, dtype=torch.bool)
+        idx = torch.arange(n - 1, device=d.device)
+        excl[idx, idx + 1] = True
+        excl[idx + 1, idx] = True
         clash = torch.relu(clash_dist - d).pow(2)
-        clash = clash.masked_fill(eye.unsqueeze(0), 0.)
-        off_diag = ~(torch.diag_embed(torch.ones_like(d[..., 0], dtype=torch.bool), offset=1)
-                     | torch.diag_embed(torch.ones_like(d[..., 0], dtype=torch.bool), offset=-1))
-        clash = clash * off_diag.to(clash.dtype)
+        clash = clash.masked_fill(excl.unsqueeze(0), 0.)
         loss = bond_loss.mean() + 0.1 * clash.mean()
         loss.backward()
         opt.step()

@@ -158,3 +158,48 @@ def test_kernel_stats_summary():
     rows = kernel_stats_summary(csvs[0], top=5)
     assert len(rows) == 5
     assert rows[0]['pct'] >= rows[1]['pct']
+
+
+@pytest.mark.timeout(300)
+def test_train_pre_script_smoke(tmp_path):
+    """train_pre.py runs a couple of tiny optimizer steps end to end."""
+    import subprocess, sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, 'train_pre.py', '--batches', '2', '--grad-accum', '1',
+         '--dim', '32', '--depth', '1', '--max-len', '24', '--msa-depth', '3',
+         '--dtype', 'fp32', '--save-every', '0', '--log-every', '1',
+         '--checkpoint', str(tmp_path / 'pre.pt')],
+        cwd=root, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert 'loss' in out.stdout
+
+
+@pytest.mark.timeout(300)
+def test_train_end2end_script_smoke(tmp_path):
+    import subprocess, sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, 'train_end2end.py', '--batches', '1',
+         '--grad-accum', '1', '--dim', '32', '--depth', '1',
+         '--structure-depth', '1', '--max-len', '16', '--msa-depth', '3',
+         '--dtype', 'fp32', '--save-every', '0', '--log-every', '1',
+         '--checkpoint', str(tmp_path / 'e2e.pt')],
+        cwd=root, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert 'loss' in out.stdout
+
+
+def test_native_relaxer():
+    import importlib.util
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    spec = importlib.util.spec_from_file_location(
+        'refinement', os.path.join(root, 'scripts', 'refinement.py'))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    coords = torch.randn(1, 20, 3) * 2
+    relaxed = mod.relax_structure(coords, iters=20)
+    d0 = (coords[:, 1:] - coords[:, :-1]).norm(dim=-1)
+    d1 = (relaxed[:, 1:] - relaxed[:, :-1]).norm(dim=-1)
+    # bond lengths move toward the 3.8 A target
+    assert (d1 - 3.8).abs().mean() < (d0 - 3.8).abs().mean()
