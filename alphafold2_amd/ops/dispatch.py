@@ -114,14 +114,14 @@ def geglu(x):
 
 
 def outer_product_mean(left, right, mask=None, eps=1e-5):
-    if using_hip(left, 'outer_mean_fwd'):
+    if left.dtype == torch.bfloat16 and using_hip(left, 'pcgemm'):
         from .hip_autograd import hip_outer_product_mean
         return hip_outer_product_mean(left, right, mask=mask, eps=eps)
     return eager.outer_product_mean(left, right, mask=mask, eps=eps)
 
 
 def triangle_mix(left, right, mix):
-    if using_hip(left, 'trimix_fwd'):
+    if left.dtype == torch.bfloat16 and using_hip(left, 'pcgemm'):
         from .hip_autograd import hip_triangle_mix
         return hip_triangle_mix(left, right, mix)
     return eager.triangle_mix(left, right, mix)

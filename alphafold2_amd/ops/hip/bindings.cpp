@@ -11,6 +11,10 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
 at::Tensor geglu_fwd(at::Tensor x);
 at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x);
 at::Tensor dist_buckets(at::Tensor coords, at::Tensor boundaries);
+at::Tensor pcgemm(at::Tensor A, at::Tensor B, long Bb, long M, long N,
+                  long K, long D,
+                  long a_bs, long a_ms, long a_ks,
+                  long b_bs, long b_ns, long b_ks, double alpha);
 at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs,
                        c10::optional<at::Tensor> rowmask);
 std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x, at::Tensor g,
@@ -36,6 +40,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward (gfx950)");
   m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward (gfx950)");
   m.def("dist_buckets", &dist_buckets, "fused cdist+bucketize (gfx950)");
+  m.def("pcgemm", &pcgemm,
+        "per-channel batched GEMM, stride-parameterized (gfx950 MFMA)");
   m.def("gatemul_fwd", &gatemul_fwd, "fused x*sigmoid(g) forward (gfx950)",
         py::arg("x"), py::arg("g"), py::arg("xs"), py::arg("gs"),
         py::arg("rowmask") = c10::nullopt);

@@ -205,9 +205,102 @@ def hip_gatemul(x, g, rowmask=None):
     return _GateMulFn.apply(x, g, rowmask)
 
 
-def hip_outer_product_mean(left, right, mask=None, eps=1e-5):
-    raise NotImplementedError
+def _pcg(A, B, M, N, K, a_m, a_k, b_n, b_k, alpha=1.0):
+    """Invoke the per-channel GEMM: C[b,m,n,d] = alpha * sum_k A*B where
+    a_m/a_k (b_n/b_k) are the AXIS INDICES (1 or 2) of A (B) that play
+    the output-m (output-n) and contraction roles."""
+    ext = _load_ext()
+    D = A.shape[-1]
+    Bb = A.shape[0]
+    return ext.pcgemm(A, B, Bb, M, N, K, D,
+                      A.stride(0), A.stride(a_m), A.stride(a_k),
+                      B.stride(0), B.stride(b_n), B.stride(b_k), alpha)
+
+
+def _pc_ok(*ts):
+    for t in ts:
+        if t.stride(-1) != 1 or t.shape[-1] % 8 != 0:
+            return False
+        if any(st % 8 != 0 for st in t.stride()[:-1]):
+            return False
+    return True
+
+
+class _TriMixFn(torch.autograd.Function):
+    """Triangle multiplicative mixing on the gfx950 per-channel GEMM.
+
+    outgoing: C[i,j] = sum_k L[i,k] R[j,k]
+    ingoing:  C[i,j] = sum_k L[k,j] R[k,i]
+    """
+
+    @staticmethod
+    def forward(ctx, left, right, mix):
+        n = left.shape[1]
+        ctx.save_for_backward(left, right)
+        ctx.mix = mix
+        if mix == 'outgoing':
+            return _pcg(left, right, n, n, n, 1, 2, 1, 2)
+        return _pcg(right, left, n, n, n, 2, 1, 2, 1)
+
+    @staticmethod
+    def backward(ctx, dC):
+        left, right = ctx.saved_tensors
+        n = left.shape[1]
+        dC = dC.contiguous()
+        if ctx.mix == 'outgoing':
+            # dL[i,k] = sum_j dC[i,j] R[j,k];  dR[j,k] = sum_i dC[i,j] L[i,k]
+            dL = _pcg(dC, right, n, n, n, 1, 2, 2, 1)
+            dR = _pcg(dC, left, n, n, n, 2, 1, 2, 1)
+        else:
+            # C[i,j] = sum_k L[k,j] R[k,i]
+            # dL[k,j] = sum_i R[k,i] dC[i,j]; dR[k,i] = sum_j L[k,j] dC[i,j]
+            dL = _pcg(right, dC, n, n, n, 1, 2, 2, 1)
+            dR = _pcg(left, dC, n, n, n, 1, 2, 1, 2)
+        return dL, dR, None
 
 
 def hip_triangle_mix(left, right, mix):
-    raise NotImplementedError
+    if not _pc_ok(left, right):
+        from . import eager
+        return eager.triangle_mix(left, right, mix)
+    return _TriMixFn.apply(left.contiguous() if left.stride(-1) != 1 else left,
+                           right, mix)
+
+
+class _OuterSumFn(torch.autograd.Function):
+    """sum_m L[b,m,i,d] R[b,m,j,d] -> (b,i,j,d), scaled by alpha."""
+
+    @staticmethod
+    def forward(ctx, left, right, alpha):
+        b, m, n, d = left.shape
+        ctx.save_for_backward(left, right)
+        ctx.alpha = alpha
+        return _pcg(left, right, n, n, m, 2, 1, 2, 1, alpha=alpha)
+
+    @staticmethod
+    def backward(ctx, dC):
+        left, right = ctx.saved_tensors
+        b, m, n, d = left.shape
+        dC = dC.contiguous()
+        # dL[m,i] = a * sum_j R[m,j] dC[i,j]; dR[m,j] = a * sum_i L[m,i] dC[i,j]
+        dL = _pcg(right, dC, m, n, n, 1, 2, 1, 2, alpha=ctx.alpha)
+        dR = _pcg(left, dC, m, n, n, 1, 2, 2, 1, alpha=ctx.alpha)
+        return dL, dR, None
+
+
+def hip_outer_product_mean(left, right, mask=None, eps=1e-5):
+    """Reference-numerics outer-product mean on the per-channel GEMM
+    (masked branch: sum_m / (m * (count + eps)) — see ops/eager.py)."""
+    m = left.shape[1]
+    if not _pc_ok(left, right):
+        from . import eager
+        return eager.outer_product_mean(left, right, mask=mask, eps=eps)
+    if mask is not None:
+        fmask = mask.to(left.dtype)
+        left = left * fmask[..., None]
+        right = right * fmask[..., None]
+        outer_sum = _OuterSumFn.apply(left.contiguous(), right.contiguous(),
+                                      1.0 / m)
+        count = torch.einsum('b m i, b m j -> b i j', fmask, fmask)
+        return outer_sum / (count[..., None] + eps)
+    return _OuterSumFn.apply(left.contiguous(), right.contiguous(), 1.0 / m)

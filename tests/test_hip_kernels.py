@@ -225,3 +225,66 @@ def test_attn_matches_model_axial_path(ext):
         m_gpu = m.cuda().bfloat16()
         out = m_gpu(x.cuda().bfloat16(), edges=edges.cuda().bfloat16())
     assert (out.float().cpu() - ref).abs().max().item() < 0.1
+
+
+# ---------------------------------------------------------------------------
+# per-channel GEMM (triangle mix + outer-product mean)
+
+
+@pytest.mark.parametrize("mix", ["outgoing", "ingoing"])
+@pytest.mark.parametrize("n", [64, 96, 100])
+def test_trimix_parity(ext, mix, n):
+    from alphafold2_amd.ops.hip_autograd import hip_triangle_mix
+    from alphafold2_amd.ops import eager
+    torch.manual_seed(0)
+    b, d = 2, 32
+    left = torch.randn(b, n, n, d, device='cuda', dtype=torch.bfloat16)
+    right = torch.randn(b, n, n, d, device='cuda', dtype=torch.bfloat16)
+    l1 = left.clone().requires_grad_(True)
+    r1 = right.clone().requires_grad_(True)
+    l2 = left.float().clone().requires_grad_(True)
+    r2 = right.float().clone().requires_grad_(True)
+
+    out1 = hip_triangle_mix(l1, r1, mix)
+    out2 = eager.triangle_mix(l2, r2, mix)
+    scale = out2.abs().max().item() + 1e-6
+    assert (out1.float() - out2).abs().max().item() < 5e-2 * scale, \
+        (out1.float() - out2).abs().max().item()
+
+    g = torch.randn_like(out2)
+    out1.backward(g.to(torch.bfloat16))
+    out2.backward(g)
+    for a1, a2, name in [(l1, l2, 'dL'), (r1, r2, 'dR')]:
+        gs = a2.grad.abs().max().item() + 1e-6
+        err = (a1.grad.float() - a2.grad).abs().max().item()
+        assert err < 5e-2 * gs, f"{mix} {name}: {err} vs scale {gs}"
+
+
+@pytest.mark.parametrize("use_mask", [False, True])
+def test_outer_mean_parity(ext, use_mask):
+    from alphafold2_amd.ops.hip_autograd import hip_outer_product_mean
+    from alphafold2_amd.ops import eager
+    torch.manual_seed(0)
+    b, m, n, d = 2, 24, 64, 32
+    left = torch.randn(b, m, n, d, device='cuda', dtype=torch.bfloat16)
+    right = torch.randn(b, m, n, d, device='cuda', dtype=torch.bfloat16)
+    mask = None
+    if use_mask:
+        mask = torch.rand(b, m, n, device='cuda') > 0.2
+    l1 = left.clone().requires_grad_(True)
+    r1 = right.clone().requires_grad_(True)
+    l2 = left.float().clone().requires_grad_(True)
+    r2 = right.float().clone().requires_grad_(True)
+
+    out1 = hip_outer_product_mean(l1, r1, mask=mask)
+    out2 = eager.outer_product_mean(l2, r2, mask=mask)
+    scale = out2.abs().max().item() + 1e-6
+    assert (out1.float() - out2).abs().max().item() < 5e-2 * scale
+
+    g = torch.randn_like(out2)
+    out1.backward(g.to(torch.bfloat16))
+    out2.backward(g)
+    for a1, a2, name in [(l1, l2, 'dL'), (r1, r2, 'dR')]:
+        gs = a2.grad.abs().max().item() + 1e-6
+        err = (a1.grad.float() - a2.grad).abs().max().item()
+        assert err < 6e-2 * max(gs, 1e-3), f"{name}: {err} vs {gs}"
