@@ -11,14 +11,13 @@
 //     operand roles/strides)
 //
 // The d axis is innermost/contiguous in every tensor; a block owns an
-// 8-channel d-chunk and a 32x32 output tile.  Staging loads 16 bytes
-// (8 channels) per (m, k) element — per-lane strided at D*2B, with the
-// sibling d-chunks of the same cacheline served by L2/L3 (inputs for a
-// whole batch entry fit the 256 MiB Infinity Cache).  LDS holds per-
-// channel [32][32] planes with +8 element row padding (2-way conflicts
-// only).  Each of 4 waves computes 2 channels with 16x16x32 bf16 MFMA.
-// The C tile round-trips through LDS so global writes are 16-byte,
-// d-contiguous.
+// 8-channel d-chunk and a 64x64 output tile (8 waves, one channel per
+// wave).  Staging loads 16 bytes (8 channels) per (m, k) element —
+// per-lane strided at D*2B, with the sibling d-chunks of the same
+// cacheline served by L2/L3 (inputs for a batch entry fit the 256 MiB
+// Infinity Cache).  LDS holds per-channel [64][32] planes with +8
+// element row padding; the C tile reuses the staging pool and drains
+// as 16-byte d-contiguous global writes.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -30,20 +29,25 @@ typedef __bf16 pb16;
 
 namespace {
 
-constexpr int TM = 32, TN = 32, TK = 32, DC = 8;
+constexpr int TM = 64, TN = 64, TK = 32, DC = 8;
+constexpr int NTHREADS = 512;               // 8 waves; one d-channel each
 constexpr int PLANE_ROW = TK + 8;           // padded row, elements
 constexpr int PLANE = TM * PLANE_ROW;       // elements per channel plane
 
-__global__ __launch_bounds__(256, 2)
+__global__ __launch_bounds__(NTHREADS, 2)
 void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
                    pb16* __restrict__ C,
                    int M, int N, int K, int D,
                    long a_bs, long a_ms, long a_ks,
                    long b_bs, long b_ns, long b_ks,
                    int ntiles, float alpha) {
-  __shared__ pb16 a_lds[DC * PLANE];
-  __shared__ pb16 b_lds[DC * PLANE];
-  __shared__ pb16 c_lds[TM * TN * DC];
+  // a/b staging planes and the C drain buffer share one pool: the
+  // C round-trip begins only after the K loop has consumed a/b
+  __shared__ pb16 pool[2 * DC * PLANE > TM * TN * DC
+                       ? 2 * DC * PLANE : TM * TN * DC];
+  pb16* a_lds = pool;
+  pb16* b_lds = pool + DC * PLANE;
+  pb16* c_lds = pool;
 
   const int mtile = blockIdx.x / ntiles;
   const int ntile = blockIdx.x - mtile * ntiles;
@@ -56,15 +60,13 @@ void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
   const pb16* Bb_ = B + (long)batch * b_bs + d0;
 
   const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
+  const int wave = threadIdx.x >> 6;   // = d channel within the chunk
 
-  f32x4_t acc[2][2][2];  // [dd][mt][nt]
+  f32x4_t acc[4][4];  // [mt][nt] 16x16 fragments of the 64x64 tile
 #pragma unroll
-  for (int dd = 0; dd < 2; ++dd)
+  for (int mt = 0; mt < 4; ++mt)
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt)
-#pragma unroll
-      for (int nt = 0; nt < 2; ++nt) acc[dd][mt][nt] = f32x4_t{0, 0, 0, 0};
+    for (int nt = 0; nt < 4; ++nt) acc[mt][nt] = f32x4_t{0, 0, 0, 0};
 
   const int m_rows = min(TM, M - m0);
   const int n_rows = min(TN, N - n0);
@@ -72,12 +74,12 @@ void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
   for (int k0 = 0; k0 < K; k0 += TK) {
     const int k_rows = min(TK, K - k0);
     __syncthreads();
-    // stage A and B tiles: each thread 4 (row, k) elements per matrix,
-    // 16B of 8 channels each, scattered into the 8 LDS planes
+    // stage A and B tiles: 64x32 (row, k) elements each, one 16-byte
+    // 8-channel load per element, scattered into the 8 LDS planes
 #pragma unroll
     for (int pass = 0; pass < 4; ++pass) {
-      int idx = threadIdx.x + pass * 256;   // 0..1023
-      int r = idx >> 5;                     // row in tile
+      int idx = threadIdx.x + pass * NTHREADS;  // 0..2047
+      int r = idx >> 5;                         // row in tile
       int kk = idx & 31;
       bf16x8_t av = {};
       bf16x8_t bv = {};
@@ -97,24 +99,21 @@ void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
     }
     __syncthreads();
 
-    // MFMA: wave handles channels 2*wave and 2*wave+1
+    // MFMA: each wave owns one d channel of the full 64x64 tile
+    const pb16* ap = a_lds + wave * PLANE;
+    const pb16* bp = b_lds + wave * PLANE;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int dd = 0; dd < 2; ++dd) {
-      const pb16* ap = a_lds + (2 * wave + dd) * PLANE;
-      const pb16* bp = b_lds + (2 * wave + dd) * PLANE;
+    for (int mt = 0; mt < 4; ++mt) {
+      // A fragment: row = mt*16 + (lane&15), k = (lane>>4)*8 + j
+      bf16x8_t af = *reinterpret_cast<const bf16x8_t*>(
+          ap + (mt * 16 + (lane & 15)) * PLANE_ROW + ((lane >> 4) << 3));
 #pragma unroll
-      for (int mt = 0; mt < 2; ++mt) {
-        // A fragment: row = mt*16 + (lane&15), k = (lane>>4)*8 + j
-        bf16x8_t af = *reinterpret_cast<const bf16x8_t*>(
-            ap + (mt * 16 + (lane & 15)) * PLANE_ROW + ((lane >> 4) << 3));
-#pragma unroll
-        for (int nt = 0; nt < 2; ++nt) {
-          bf16x8_t bf = *reinterpret_cast<const bf16x8_t*>(
-              bp + (nt * 16 + (lane & 15)) * PLANE_ROW + ((lane >> 4) << 3));
-          acc[dd][mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af, bf, acc[dd][mt][nt], 0, 0, 0);
-        }
+      for (int nt = 0; nt < 4; ++nt) {
+        bf16x8_t bf = *reinterpret_cast<const bf16x8_t*>(
+            bp + (nt * 16 + (lane & 15)) * PLANE_ROW + ((lane >> 4) << 3));
+        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af, bf, acc[mt][nt], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
@@ -122,29 +121,26 @@ void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
 
   // C tile -> LDS in [m][n][d] layout, then 16B coalesced global writes
   __syncthreads();
+  const int d = wave;
 #pragma unroll
-  for (int dd = 0; dd < 2; ++dd) {
-    const int d = 2 * wave + dd;
+  for (int mt = 0; mt < 4; ++mt)
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt)
+    for (int nt = 0; nt < 4; ++nt)
 #pragma unroll
-      for (int nt = 0; nt < 2; ++nt)
-#pragma unroll
-        for (int reg = 0; reg < 4; ++reg) {
-          const int row = mt * 16 + (lane >> 4) * 4 + reg;
-          const int col = nt * 16 + (lane & 15);
-          c_lds[(row * TN + col) * DC + d] =
-              (pb16)(acc[dd][mt][nt][reg] * alpha);
-        }
-  }
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = mt * 16 + (lane >> 4) * 4 + reg;
+        const int col = nt * 16 + (lane & 15);
+        c_lds[(row * TN + col) * DC + d] =
+            (pb16)(acc[mt][nt][reg] * alpha);
+      }
   __syncthreads();
-  // 32*32 (m,n) cells * 16B = 1024 chunks; 4 per thread
+  // 64*64 (m,n) cells * 16B = 4096 chunks; 8 per thread
   pb16* Cb = C + (((long)batch * M) * N) * D + d0;
 #pragma unroll
-  for (int pass = 0; pass < 4; ++pass) {
-    int idx = threadIdx.x + pass * 256;
-    int r = idx >> 5;
-    int cc = idx & 31;
+  for (int pass = 0; pass < 8; ++pass) {
+    int idx = threadIdx.x + pass * NTHREADS;
+    int r = idx >> 6;
+    int cc = idx & 63;
     if (r < m_rows && cc < n_rows) {
       *reinterpret_cast<bf16x8_t*>(
           Cb + ((long)(m0 + r) * N + (n0 + cc)) * D) =
@@ -168,7 +164,7 @@ at::Tensor pcgemm(at::Tensor A, at::Tensor B, long Bb, long M, long N,
   const int ntiles = (N + TN - 1) / TN;
   dim3 grid(mtiles * ntiles, Bb * (D / DC));
   auto stream = at::cuda::getCurrentHIPStream();
-  hipLaunchKernelGGL(pcgemm_kernel, grid, dim3(256), 0, stream,
+  hipLaunchKernelGGL(pcgemm_kernel, grid, dim3(NTHREADS), 0, stream,
                      reinterpret_cast<const pb16*>(A.data_ptr()),
                      reinterpret_cast<const pb16*>(B.data_ptr()),
                      reinterpret_cast<pb16*>(C.data_ptr()),
