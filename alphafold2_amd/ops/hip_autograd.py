@@ -131,7 +131,12 @@ class _AttentionPackedFn(torch.autograd.Function):
         B, L, W = packed.shape
         dh = inner // heads
 
-        dpacked = torch.zeros_like(packed)
+        # dq/dk/dv slices are fully overwritten by the kernel; only the
+        # remaining channels (the gate block) must start at zero — their
+        # gradient arrives via the other autograd path and is summed
+        dpacked = torch.empty_like(packed)
+        if W > 3 * inner:
+            dpacked.narrow(-1, 3 * inner, W - 3 * inner).zero_()
 
         def view(t, off):
             return t.narrow(-1, off, inner)                 .view(B, L, heads, dh).permute(0, 2, 1, 3)
