@@ -148,6 +148,60 @@ __device__ __forceinline__ void stage_tile_t(const bf16_t* __restrict__ g,
   }
 }
 
+// async-stage split (guide T14): issue the tile's global loads into
+// registers early (overlapping prior compute), write LDS after the
+// barrier.  The same registers can be stored in BOTH layouts (normal +
+// transposed), halving global traffic for double-layout tiles.
+struct StageRegs {
+  float4 v0, v1;
+};
+
+__device__ __forceinline__ void stage_load(const bf16_t* __restrict__ g,
+                                           long row_stride, int rows,
+                                           StageRegs& r) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    int idx = tid + pass * 256;
+    int row = idx >> 3;
+    int c16 = (idx & 7) << 4;
+    float4 val = {0, 0, 0, 0};
+    if (row < rows) {
+      val = *reinterpret_cast<const float4*>(
+          reinterpret_cast<const char*>(g + row * row_stride) + c16);
+    }
+    (pass ? r.v1 : r.v0) = val;
+  }
+}
+
+__device__ __forceinline__ void stage_store(const StageRegs& r, char* lds) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    int idx = tid + pass * 256;
+    int row = idx >> 3;
+    int c16 = (idx & 7) << 4;
+    *reinterpret_cast<float4*>(lds + swz(row, c16)) = (pass ? r.v1 : r.v0);
+  }
+}
+
+__device__ __forceinline__ void stage_store_t(const StageRegs& r, char* lds) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    int idx = tid + pass * 256;
+    int row = idx >> 3;
+    int c8 = (idx & 7) << 3;  // first of 8 source cols
+    const float4 val = pass ? r.v1 : r.v0;
+    const bf16_t* vv = reinterpret_cast<const bf16_t*>(&val);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      *reinterpret_cast<bf16_t*>(
+          lds + swz(c8 + j, row * (int)sizeof(bf16_t))) = vv[j];
+    }
+  }
+}
+
 // read an 8-bf16 A/B fragment (k = (lane>>4)*8 + j) for tile row `row`,
 // k-block `kblk` (32 wide) from a swizzled LDS tile
 __device__ __forceinline__ bf16x8 frag_row(const char* lds, int row,
@@ -244,16 +298,25 @@ void attn_fwd_kernel(TView q, TView k, TView v,
   }
 
   const int n_kv = (Lk + BK - 1) / BK;
+  StageRegs kreg, vreg;
+  stage_load(k_g, k.rs, min(BK, Lk), kreg);
+  stage_load(v_g, v.rs, min(BK, Lk), vreg);
   for (int t = 0; t < n_kv; ++t) {
     const int kv_rows = min(BK, Lk - t * BK);
     __syncthreads();
-    stage_tile(k_g + (long)t * BK * k.rs, k.rs, kv_rows, k_lds);
-    stage_tile_t(v_g + (long)t * BK * v.rs, v.rs, kv_rows, vt_lds);
+    stage_store(kreg, k_lds);
+    stage_store_t(vreg, vt_lds);
     if (HAS_MASK && threadIdx.x < BK) {
       m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
           ? mask[(long)batch * Lk + t * BK + threadIdx.x] : 0;
     }
     __syncthreads();
+    if (t + 1 < n_kv) {
+      // next tile's loads fly while this tile computes
+      const int next_rows = min(BK, Lk - (t + 1) * BK);
+      stage_load(k_g + (long)(t + 1) * BK * k.rs, k.rs, next_rows, kreg);
+      stage_load(v_g + (long)(t + 1) * BK * v.rs, v.rs, next_rows, vreg);
+    }
 
     // S = Q K^T  (16 q x 64 kv per wave); setprio favors the MFMA
     // cluster when co-resident waves are staging (guide T5)
@@ -457,17 +520,25 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
   for (int c = 0; c < 4; ++c) dq_acc[c] = f32x4{0, 0, 0, 0};
 
   const int n_kv = (Lk + BK - 1) / BK;
+  StageRegs kreg, vreg;
+  stage_load(k_g, k.rs, min(BK, Lk), kreg);
+  stage_load(v_g, v.rs, min(BK, Lk), vreg);
   for (int t = 0; t < n_kv; ++t) {
     const int kv_rows = min(BK, Lk - t * BK);
     __syncthreads();
-    stage_tile(k_g + (long)t * BK * k.rs, k.rs, kv_rows, k_lds);
-    stage_tile_t(k_g + (long)t * BK * k.rs, k.rs, kv_rows, kt_lds);
-    stage_tile(v_g + (long)t * BK * v.rs, v.rs, kv_rows, v_lds);
+    stage_store(kreg, k_lds);
+    stage_store_t(kreg, kt_lds);  // one load feeds both layouts
+    stage_store(vreg, v_lds);
     if (HAS_MASK && threadIdx.x < BK) {
       m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
           ? mask[(long)batch * Lk + t * BK + threadIdx.x] : 0;
     }
     __syncthreads();
+    if (t + 1 < n_kv) {
+      const int next_rows = min(BK, Lk - (t + 1) * BK);
+      stage_load(k_g + (long)(t + 1) * BK * k.rs, k.rs, next_rows, kreg);
+      stage_load(v_g + (long)(t + 1) * BK * v.rs, v.rs, next_rows, vreg);
+    }
 
     // recompute S then P = exp(S*scale + bias - lse)
     f32x4 p[4], dp[4];
@@ -645,13 +716,16 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
   }
 
   const int n_q = (Lq + BQ - 1) / BQ;
+  StageRegs qreg, doreg;
+  stage_load(q_g, q.rs, min(BQ, Lq), qreg);
+  stage_load(do_g, dout.rs, min(BQ, Lq), doreg);
   for (int t = 0; t < n_q; ++t) {
     const int q_rows = min(BQ, Lq - t * BQ);
     __syncthreads();
-    stage_tile(q_g + (long)t * BQ * q.rs, q.rs, q_rows, q_lds);
-    stage_tile_t(q_g + (long)t * BQ * q.rs, q.rs, q_rows, qt_lds);
-    stage_tile(do_g + (long)t * BQ * dout.rs, dout.rs, q_rows, do_lds);
-    stage_tile_t(do_g + (long)t * BQ * dout.rs, dout.rs, q_rows, dot_lds);
+    stage_store(qreg, q_lds);
+    stage_store_t(qreg, qt_lds);    // one load feeds both layouts
+    stage_store(doreg, do_lds);
+    stage_store_t(doreg, dot_lds);
     if (HAS_BIAS) {
       stage_tile_rowstride(bias_g + (long)t * BQ * Lk + (long)ktile * BK,
                            Lk, q_rows, min(BK, Lk - ktile * BK), b_lds);
@@ -662,6 +736,12 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
       delta_lds[threadIdx.x] = (threadIdx.x < q_rows) ? delta_g[qq] : 0.f;
     }
     __syncthreads();
+    if (t + 1 < n_q) {
+      const int next_rows = min(BQ, Lq - (t + 1) * BQ);
+      stage_load(q_g + (long)(t + 1) * BQ * q.rs, q.rs, next_rows, qreg);
+      stage_load(do_g + (long)(t + 1) * BQ * dout.rs, dout.rs, next_rows,
+                 doreg);
+    }
 
     // S^T = K Q^T : A = K (k = d), B col = q row of Q (k = d contiguous)
     f32x4 pt[4], dpt[4];
