@@ -87,8 +87,13 @@ def main():
     use_graph = (device.type == 'cuda' and not args.no_graph
                  and (world_size == 1
                       or os.environ.get('AF2AMD_GRAPH_DDP') == '1'))
-    optimizer = torch.optim.Adam(model.parameters(), lr=3e-4,
-                                 capturable=use_graph, foreach=True)
+    try:
+        # single fused multi-tensor Adam kernel (ROCm-supported)
+        optimizer = torch.optim.Adam(model.parameters(), lr=3e-4,
+                                     capturable=use_graph, fused=True)
+    except (RuntimeError, ValueError):
+        optimizer = torch.optim.Adam(model.parameters(), lr=3e-4,
+                                     capturable=use_graph, foreach=True)
 
     use_bf16 = args.dtype == 'bf16' and device.type == 'cuda'
 
