@@ -28,8 +28,8 @@ def parse_args():
     p.add_argument('--depth', type=int, default=12)
     p.add_argument('--crop-len', type=int, default=256)
     p.add_argument('--msa-depth', type=int, default=128)
-    p.add_argument('--batch', type=int, default=4,
-                   help='per-GPU batch size (4 fills ~140 GB of the '
+    p.add_argument('--batch', type=int, default=5,
+                   help='per-GPU batch size (5 fills ~175 GB of the '
                         '288 GB HBM3E without activation checkpointing)')
     p.add_argument('--heads', type=int, default=8)
     p.add_argument('--dim-head', type=int, default=64)
