@@ -203,3 +203,35 @@ def test_native_relaxer():
     d1 = (relaxed[:, 1:] - relaxed[:, :-1]).norm(dim=-1)
     # bond lengths move toward the 3.8 A target
     assert (d1 - 3.8).abs().mean() < (d0 - 3.8).abs().mean()
+
+
+@pytest.mark.timeout(300)
+def test_predict_cli(tmp_path):
+    """predict.py: sequence in -> CA-trace PDB with confidence out."""
+    import subprocess, sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out_pdb = str(tmp_path / 'pred.pdb')
+    r = subprocess.run(
+        [sys.executable, 'predict.py', '--seq', 'MKTAYIAKQRQISFVKSHFSRQ',
+         '--dim', '32', '--depth', '1', '--recycles', '1', '--relax',
+         '--out', out_pdb],
+        cwd=root, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-2000:]
+    text = open(out_pdb).read()
+    assert text.count('ATOM') == 22
+    assert 'CA' in text
+
+
+@pytest.mark.timeout(300)
+def test_predict_cli_a3m(tmp_path):
+    import subprocess, sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    a3m = tmp_path / 'q.a3m'
+    a3m.write_text('>q\nACDEFGHIKLMNPQRST\n>h1\nACDEFGHIKLMNPQRST\n')
+    out_pdb = str(tmp_path / 'p.pdb')
+    r = subprocess.run(
+        [sys.executable, 'predict.py', '--a3m', str(a3m), '--dim', '32',
+         '--depth', '1', '--recycles', '1', '--out', out_pdb],
+        cwd=root, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(out_pdb)
