@@ -202,6 +202,64 @@ __device__ __forceinline__ void stage_store_t(const StageRegs& r, char* lds) {
   }
 }
 
+// block-width/row-count generic staging (the fwd kernel runs 512
+// threads over a 128-row Q tile and 64-row K/V tiles)
+template <int NT, int R>
+struct StageRegsN {
+  float4 v[(R * 8 + NT - 1) / NT];
+};
+
+template <int NT, int R>
+__device__ __forceinline__ void stage_load_n(const bf16_t* __restrict__ g,
+                                             long row_stride, int rows,
+                                             StageRegsN<NT, R>& r) {
+  constexpr int P = (R * 8 + NT - 1) / NT;
+#pragma unroll
+  for (int pass = 0; pass < P; ++pass) {
+    int idx = threadIdx.x + pass * NT;
+    int row = idx >> 3;
+    int c16 = (idx & 7) << 4;
+    float4 val = {0, 0, 0, 0};
+    if (row < rows) {
+      val = *reinterpret_cast<const float4*>(
+          reinterpret_cast<const char*>(g + row * row_stride) + c16);
+    }
+    r.v[pass] = val;
+  }
+}
+
+template <int NT, int R>
+__device__ __forceinline__ void stage_store_n(const StageRegsN<NT, R>& r,
+                                              char* lds) {
+  constexpr int P = (R * 8 + NT - 1) / NT;
+#pragma unroll
+  for (int pass = 0; pass < P; ++pass) {
+    int idx = threadIdx.x + pass * NT;
+    int row = idx >> 3;
+    int c16 = (idx & 7) << 4;
+    *reinterpret_cast<float4*>(lds + swz(row, c16)) = r.v[pass];
+  }
+}
+
+template <int NT, int R>
+__device__ __forceinline__ void stage_store_t_n(const StageRegsN<NT, R>& r,
+                                                char* lds) {
+  constexpr int P = (R * 8 + NT - 1) / NT;
+#pragma unroll
+  for (int pass = 0; pass < P; ++pass) {
+    int idx = threadIdx.x + pass * NT;
+    int row = idx >> 3;
+    int c8 = (idx & 7) << 3;
+    const float4 val = r.v[pass];
+    const bf16_t* vv = reinterpret_cast<const bf16_t*>(&val);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      *reinterpret_cast<bf16_t*>(
+          lds + swz(c8 + j, row * (int)sizeof(bf16_t))) = vv[j];
+    }
+  }
+}
+
 // read an 8-bf16 A/B fragment (k = (lane>>4)*8 + j) for tile row `row`,
 // k-block `kblk` (32 wide) from a swizzled LDS tile
 __device__ __forceinline__ bf16x8 frag_row(const char* lds, int row,
@@ -246,18 +304,21 @@ __device__ __forceinline__ float rowred_sum(const f32x4 s[4], int reg) {
   return v;
 }
 
+constexpr int FBQ = 128;    // fwd query rows per workgroup
+constexpr int FNT = 512;    // fwd threads (8 waves x 16 q rows)
+
 template <bool HAS_BIAS, bool HAS_MASK>
-__global__ __launch_bounds__(256, 2)
+__global__ __launch_bounds__(FNT, 2)
 void attn_fwd_kernel(TView q, TView k, TView v,
                      const bf16_t* __restrict__ bias,
                      const unsigned char* __restrict__ mask,
                      TViewMut out, float* __restrict__ lse,
                      int Lq, int Lk, int heads, int bias_repeat,
                      float scale) {
-  __shared__ char q_lds[BQ * ROWB];
+  __shared__ char q_lds[FBQ * ROWB];
   __shared__ char k_lds[BK * ROWB];
   __shared__ char vt_lds[BK * ROWB];  // V transposed: [dv][kv]
-  __shared__ char p_lds[NWAVES][16 * ROWB];
+  __shared__ char p_lds[FNT / 64][16 * ROWB];
   __shared__ unsigned char m_lds[BK];
 
   const int qtile = blockIdx.x;
@@ -267,18 +328,22 @@ void attn_fwd_kernel(TView q, TView k, TView v,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  const bf16_t* q_g = q.base(batch, head) + (long)qtile * BQ * q.rs;
+  const bf16_t* q_g = q.base(batch, head) + (long)qtile * FBQ * q.rs;
   const bf16_t* k_g = k.base(batch, head);
   const bf16_t* v_g = v.base(batch, head);
   const bf16_t* bias_g = nullptr;
   if (HAS_BIAS) {
     const int bias_batch = batch / bias_repeat;
     bias_g = bias + ((long)(bias_batch * heads + head) * Lq
-                     + (long)qtile * BQ) * Lk;
+                     + (long)qtile * FBQ) * Lk;
   }
 
-  const int q_rows = min(BQ, Lq - qtile * BQ);
-  stage_tile(q_g, q.rs, q_rows, q_lds);
+  const int q_rows = min(FBQ, Lq - qtile * FBQ);
+  {
+    StageRegsN<FNT, FBQ> qr;
+    stage_load_n<FNT, FBQ>(q_g, q.rs, q_rows, qr);
+    stage_store_n<FNT, FBQ>(qr, q_lds);
+  }
   __syncthreads();
 
   // per-wave Q fragments (rows wave*16 + (lane&15))
@@ -298,14 +363,14 @@ void attn_fwd_kernel(TView q, TView k, TView v,
   }
 
   const int n_kv = (Lk + BK - 1) / BK;
-  StageRegs kreg, vreg;
-  stage_load(k_g, k.rs, min(BK, Lk), kreg);
-  stage_load(v_g, v.rs, min(BK, Lk), vreg);
+  StageRegsN<FNT, BK> kreg, vreg;
+  stage_load_n<FNT, BK>(k_g, k.rs, min(BK, Lk), kreg);
+  stage_load_n<FNT, BK>(v_g, v.rs, min(BK, Lk), vreg);
   for (int t = 0; t < n_kv; ++t) {
     const int kv_rows = min(BK, Lk - t * BK);
     __syncthreads();
-    stage_store(kreg, k_lds);
-    stage_store_t(vreg, vt_lds);
+    stage_store_n<FNT, BK>(kreg, k_lds);
+    stage_store_t_n<FNT, BK>(vreg, vt_lds);
     if (HAS_MASK && threadIdx.x < BK) {
       m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
           ? mask[(long)batch * Lk + t * BK + threadIdx.x] : 0;
@@ -314,8 +379,10 @@ void attn_fwd_kernel(TView q, TView k, TView v,
     if (t + 1 < n_kv) {
       // next tile's loads fly while this tile computes
       const int next_rows = min(BK, Lk - (t + 1) * BK);
-      stage_load(k_g + (long)(t + 1) * BK * k.rs, k.rs, next_rows, kreg);
-      stage_load(v_g + (long)(t + 1) * BK * v.rs, v.rs, next_rows, vreg);
+      stage_load_n<FNT, BK>(k_g + (long)(t + 1) * BK * k.rs, k.rs,
+                            next_rows, kreg);
+      stage_load_n<FNT, BK>(v_g + (long)(t + 1) * BK * v.rs, v.rs,
+                            next_rows, vreg);
     }
 
     // S = Q K^T  (16 q x 64 kv per wave); setprio favors the MFMA
@@ -392,6 +459,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
       p_frag[kblk] = frag_row(pw, lane & 15, kblk);
 
     // O += P V  (B-operand = V^T rows: one b128 read per fragment)
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       f32x4 acc = o_acc[c];
@@ -403,11 +471,12 @@ void attn_fwd_kernel(TView q, TView k, TView v,
       }
       o_acc[c] = acc;
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // epilogue: O /= l, store out + lse
-  bf16_t* out_g = out.base(batch, head) + (long)qtile * BQ * out.rs;
-  float* lse_g = lse + (long)bh * Lq + (long)qtile * BQ;
+  bf16_t* out_g = out.base(batch, head) + (long)qtile * FBQ * out.rs;
+  float* lse_g = lse + (long)bh * Lq + (long)qtile * FBQ;
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
     const int row = wave * 16 + (lane >> 4) * 4 + reg;
@@ -934,13 +1003,13 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   if (has_bias) bias_c = bias->contiguous();
   if (has_mask) mask_u8 = mask->to(at::kByte).contiguous();
 
-  dim3 grid((Lq + BQ - 1) / BQ, B * H);
+  dim3 grid((Lq + FBQ - 1) / FBQ, B * H);
   auto stream = at::cuda::getCurrentHIPStream();
   TView qv = make_view(q), kv = make_view(k), vv = make_view(v);
   TViewMut ov = make_view_mut(out);
 
 #define DISPATCH(HB, HM)                                                      \
-  hipLaunchKernelGGL((attn_fwd_kernel<HB, HM>), grid, dim3(256), 0, stream,   \
+  hipLaunchKernelGGL((attn_fwd_kernel<HB, HM>), grid, dim3(FNT), 0, stream,   \
       qv, kv, vv,                                                             \
       has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
       has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
