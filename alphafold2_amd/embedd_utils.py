@@ -97,3 +97,19 @@ def get_t5_embedd(seq, tokenizer, encoder, msa_data=None, device=None):
     token_reps = embedding.last_hidden_state[:, shift_left:shift_right].to(device)
     token_reps = expand_dims_to(token_reps, 4 - len(token_reps.shape))
     return token_reps.float()
+
+
+def get_all_protein_ids(dataloader, verbose=False):
+    """Collect the protein-entry ids from a sidechainnet-style
+    dataloader (fixed version of the reference helper, which read an
+    undefined global and never advanced its iterator correctly)."""
+    ids = set()
+    for batch in dataloader:
+        for pid in getattr(batch, 'pids', []):
+            max_len_10 = len(pid) < 10
+            fragments = [len(x) <= 4 for x in pid.split("_")]
+            if max_len_10 and all(fragments):
+                ids.add(pid)
+            elif verbose:
+                print("skip:", pid)
+    return ids

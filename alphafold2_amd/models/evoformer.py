@@ -98,6 +98,7 @@ class Attention(nn.Module):
         nn.init.constant_(self.gating.weight, 0.)
         nn.init.constant_(self.gating.bias, 1.)
 
+        self.dropout_p = dropout
         self.dropout = nn.Dropout(dropout)
         init_zero_(self.to_out)
 
@@ -120,7 +121,7 @@ class Attention(nn.Module):
             inner = self.to_q.weight.shape[0]
             fused = F.linear(x, w, bias_cat)
 
-            if tie_dim is None:
+            if tie_dim is None and not (self.dropout_p > 0 and self.training):
                 # packed fast path: the fused kernel consumes q/k/v as
                 # strided slices and writes one packed grad (no
                 # split-backward concatenation)
@@ -150,7 +151,8 @@ class Attention(nn.Module):
 
         out = ops.attention_core(
             q, k, v, bias=attn_bias, mask=mask, context_mask=context_mask,
-            tie_dim=tie_dim, bias_repeat=attn_bias_repeat)
+            tie_dim=tie_dim, bias_repeat=attn_bias_repeat,
+            dropout=self.dropout_p, training=self.training)
 
         out = out.transpose(-2, -3).reshape(*x.shape[:-1], -1)
 

@@ -64,12 +64,15 @@ def using_hip(t: torch.Tensor, opname: str) -> bool:
 
 
 def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
-                   tie_dim=None, bias_repeat=1):
+                   tie_dim=None, bias_repeat=1, dropout=0., training=False):
     """bias (when given) has shape (B // bias_repeat, h, Lq, Lk); the
     repeat fold is resolved inside the fused kernel (never materialized),
-    and expanded explicitly only on the eager path."""
+    and expanded explicitly only on the eager path.  Attention-prob
+    dropout (reference alphafold2.py:172) routes to the eager path."""
+    drop_active = dropout > 0. and training
     fusable = (
         tie_dim is None
+        and not drop_active
         and q.dtype == torch.bfloat16
         and q.shape[-1] == 64
         and (bias is None or bias.dtype == torch.bfloat16)
@@ -83,7 +86,8 @@ def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
     if bias is not None and bias_repeat != 1:
         bias = bias.repeat_interleave(bias_repeat, dim=0)
     return eager.attention_core(q, k, v, bias=bias, mask=mask,
-                                context_mask=context_mask, tie_dim=tie_dim)
+                                context_mask=context_mask, tie_dim=tie_dim,
+                                dropout=dropout, training=training)
 
 
 def attention_core_packed(packed, heads, inner, bias=None, mask=None,

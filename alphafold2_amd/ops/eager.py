@@ -10,7 +10,7 @@ import torch.nn.functional as F
 
 
 def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
-                   tie_dim=None):
+                   tie_dim=None, dropout=0., training=False):
     """softmax(q k^T * scale + bias) v with optional key masking.
 
     * q, k, v: (B, h, n, d) — B may fold extra axes (MSA rows, pair rows)
@@ -45,6 +45,8 @@ def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
         dots = dots.masked_fill(~full, mask_value)
 
     attn = dots.softmax(dim=-1)
+    if dropout > 0. and training:
+        attn = F.dropout(attn, p=dropout, training=True)
     return torch.einsum('b h i j, b h j d -> b h i d', attn, v)
 
 

@@ -26,4 +26,5 @@ from .geometry import (  # noqa: F401
 from .embedd_utils import (  # noqa: F401
     ids_to_embed_input, ids_to_prottran_input,
     get_prottran_embedd, get_msa_embedd, get_esm_embedd, get_t5_embedd,
+    get_all_protein_ids,
 )

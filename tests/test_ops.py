@@ -152,3 +152,19 @@ def test_hip_attention_parity(dtype):
                                bias=bias.float())
     tol = 2e-2 if dtype == torch.bfloat16 else 1e-4
     assert (out.float() - ref).abs().max().item() < tol
+
+
+def test_attention_dropout_eager():
+    """attn_dropout routes through the eager path and actually drops."""
+    torch.manual_seed(0)
+    from alphafold2_amd.models.evoformer import Attention
+    m = Attention(dim=32, heads=2, dim_head=16, dropout=0.5).train()
+    x = torch.randn(2, 8, 32)
+    o1 = m(x)
+    o2 = m(x)
+    assert not torch.allclose(o1, o2), 'dropout must be stochastic'
+    m.eval()
+    with torch.no_grad():
+        o3 = m(x)
+        o4 = m(x)
+    assert torch.allclose(o3, o4), 'eval must be deterministic'
