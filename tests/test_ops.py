@@ -159,6 +159,7 @@ def test_attention_dropout_eager():
     torch.manual_seed(0)
     from alphafold2_amd.models.evoformer import Attention
     m = Attention(dim=32, heads=2, dim_head=16, dropout=0.5).train()
+    torch.nn.init.normal_(m.to_out.weight)  # zero-init would hide dropout
     x = torch.randn(2, 8, 32)
     o1 = m(x)
     o2 = m(x)
