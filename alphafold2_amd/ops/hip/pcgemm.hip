@@ -40,7 +40,7 @@ void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
                    int M, int N, int K, int D,
                    long a_bs, long a_ms, long a_ks,
                    long b_bs, long b_ns, long b_ks,
-                   int ntiles, float alpha) {
+                   int ntiles, int mtiles, int swizzle, float alpha) {
   // a/b staging planes and the C drain buffer share one pool: the
   // C round-trip begins only after the K loop has consumed a/b
   __shared__ pb16 pool[2 * DC * PLANE > TM * TN * DC
@@ -49,12 +49,31 @@ void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
   pb16* b_lds = pool + DC * PLANE;
   pb16* c_lds = pool;
 
-  const int mtile = blockIdx.x / ntiles;
-  const int ntile = blockIdx.x - mtile * ntiles;
-  const int m0 = mtile * TM, n0 = ntile * TN;
+  // logical work id L enumerates (batch, tile, chunk) with the d-chunk
+  // fastest; the XCD swizzle places the 8 sibling chunks of a tile
+  // (which share 128-byte cachelines of A/B/C) on ONE XCD's L2 —
+  // without it each sibling re-fetches the same line into a different
+  // XCD L2 (8x HBM/L3 traffic; MfmaUtil measured at 3%).
+  // phys = (G%8) + 8*(m + 8*(G/8)) for L = 8G + m  (bijective when the
+  // grid is a multiple of 64; identity otherwise).
+  long L = (long)blockIdx.y * gridDim.x + blockIdx.x;
+  if (swizzle) {
+    const long phys = L;
+    const long c = phys & 7;
+    const long t = phys >> 3;
+    const long m_ = t & 7;
+    const long Ghi = t >> 3;
+    L = ((Ghi << 3) + c) * 8 + m_;
+  }
   const int dchunks = D / DC;
-  const int batch = blockIdx.y / dchunks;
-  const int d0 = (blockIdx.y - batch * dchunks) * DC;
+  const int chunk = L % dchunks;
+  const long Lt = L / dchunks;
+  const int tile = Lt % ((long)mtiles * ntiles);
+  const int batch = Lt / ((long)mtiles * ntiles);
+  const int mtile = tile / ntiles;
+  const int ntile = tile - mtile * ntiles;
+  const int m0 = mtile * TM, n0 = ntile * TN;
+  const int d0 = chunk * DC;
 
   const pb16* Ab = A + (long)batch * a_bs + d0;
   const pb16* Bb_ = B + (long)batch * b_bs + d0;
@@ -162,6 +181,8 @@ at::Tensor pcgemm(at::Tensor A, at::Tensor B, long Bb, long M, long N,
   auto C = at::empty({Bb, M, N, D}, A.options());
   const int mtiles = (M + TM - 1) / TM;
   const int ntiles = (N + TN - 1) / TN;
+  const long nblocks = (long)mtiles * ntiles * Bb * (D / DC);
+  const int swizzle = (D / DC) % 8 == 0 && nblocks % 64 == 0;
   dim3 grid(mtiles * ntiles, Bb * (D / DC));
   auto stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(pcgemm_kernel, grid, dim3(NTHREADS), 0, stream,
@@ -170,6 +191,6 @@ at::Tensor pcgemm(at::Tensor A, at::Tensor B, long Bb, long M, long N,
                      reinterpret_cast<pb16*>(C.data_ptr()),
                      (int)M, (int)N, (int)K, (int)D,
                      a_bs, a_ms, a_ks, b_bs, b_ns, b_ks,
-                     ntiles, (float)alpha);
+                     ntiles, mtiles, swizzle, (float)alpha);
   return C;
 }
