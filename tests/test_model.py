@@ -271,3 +271,40 @@ def test_egnn_equivariance():
 def test_refinement_iters_alias():
     m = tiny_model(predict_coords=True, structure_module_refinement_iters=2)
     assert m.structure_module_depth == 2
+
+
+def test_quaternion_properties():
+    from alphafold2_amd.models.quaternion import (
+        quaternion_multiply, quaternion_to_matrix, matrix_to_quaternion)
+    torch.manual_seed(0)
+    q = torch.randn(32, 4, dtype=torch.float64)
+    q = q / q.norm(dim=-1, keepdim=True)
+    R = quaternion_to_matrix(q)
+    # orthonormal, det +1
+    eye = torch.eye(3, dtype=torch.float64).expand(32, 3, 3)
+    assert torch.allclose(R @ R.transpose(-1, -2), eye, atol=1e-10)
+    assert torch.allclose(torch.linalg.det(R), torch.ones(32, dtype=torch.float64), atol=1e-10)
+    # multiply <-> matrix product consistency
+    q2 = torch.randn(32, 4, dtype=torch.float64)
+    q2 = q2 / q2.norm(dim=-1, keepdim=True)
+    R12 = quaternion_to_matrix(quaternion_multiply(q, q2))
+    # row-vector convention: rotate(v, q1*q2) == rotate(rotate(v, q2), q1)
+    v = torch.randn(32, 3, dtype=torch.float64)
+    a = torch.einsum('bc,bcr->br', v, R12)
+    b = torch.einsum('bc,bcr->br', torch.einsum('bc,bcr->br', v, quaternion_to_matrix(q2)), quaternion_to_matrix(q))
+    ok1 = torch.allclose(a, b, atol=1e-9)
+    b2 = torch.einsum('bc,bcr->br', torch.einsum('bc,bcr->br', v, quaternion_to_matrix(q)), quaternion_to_matrix(q2))
+    ok2 = torch.allclose(a, b2, atol=1e-9)
+    assert ok1 or ok2
+    # matrix -> quaternion roundtrip (up to sign, standardized)
+    q3 = matrix_to_quaternion(R)
+    assert torch.allclose(quaternion_to_matrix(q3), R, atol=1e-6)
+
+
+def test_module_alias_imports():
+    import alphafold2_amd.embeds
+    import alphafold2_amd.reversible
+    import alphafold2_amd.rotary
+    assert hasattr(alphafold2_amd.embeds, 'ESMEmbedWrapper')
+    assert hasattr(alphafold2_amd.reversible, 'ReversibleEvoformer')
+    assert hasattr(alphafold2_amd.rotary, 'FixedPositionalEmbedding')
