@@ -403,6 +403,15 @@ void attn_fwd_kernel(TView q, TView k, TView v,
     __builtin_amdgcn_s_setprio(0);
 
     // scale + bias + masking (C layout: row=(lane>>4)*4+reg, col=lane&15)
+    const bf16_t* brow[4];
+    if (HAS_BIAS) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = wave * 16 + (lane >> 4) * 4 + reg;
+        brow[reg] = (row < q_rows)
+            ? bias_g + (long)row * Lk + t * BK : nullptr;
+      }
+    }
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       const int col = c * 16 + (lane & 15);
@@ -411,11 +420,8 @@ void attn_fwd_kernel(TView q, TView k, TView v,
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         float val = s[c][reg] * scale;
-        if (HAS_BIAS && col_ok) {
-          const int row = wave * 16 + (lane >> 4) * 4 + reg;
-          if (row < q_rows)
-            val += to_f32(bias_g[(long)row * Lk + t * BK + col]);
-        }
+        if (HAS_BIAS && col_ok && brow[reg] != nullptr)
+          val += to_f32(brow[reg][col]);
         s[c][reg] = col_ok ? val : NEG_INF;
       }
     }
@@ -622,6 +628,15 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
       }
       p[c] = acc;
     }
+    const bf16_t* brow[4];
+    if (HAS_BIAS) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = wave * 16 + (lane >> 4) * 4 + reg;
+        brow[reg] = (row < q_rows)
+            ? bias_g + (long)row * Lk + t * BK : nullptr;
+      }
+    }
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       const int col = c * 16 + (lane & 15);
@@ -629,9 +644,8 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         float val = p[c][reg] * scale;
-        const int row = wave * 16 + (lane >> 4) * 4 + reg;
-        if (HAS_BIAS && col_ok && row < q_rows)
-          val += to_f32(bias_g[(long)row * Lk + t * BK + col]);
+        if (HAS_BIAS && col_ok && brow[reg] != nullptr)
+          val += to_f32(brow[reg][col]);
         p[c][reg] = (col_ok && lse_r[reg] > NEG_INF)
             ? __expf(val - lse_r[reg]) : 0.f;
       }
