@@ -730,6 +730,7 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
                          const float* __restrict__ delta,
                          TViewMut dk, TViewMut dv,
                          float* __restrict__ dbias,
+                         int dbias_chunks, long dbias_stride,
                          int Lq, int Lk, int heads, int bias_repeat,
                          float scale) {
   __shared__ char k_lds[BK * ROWB];
@@ -762,8 +763,13 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
   const int bias_batch = batch / bias_repeat;
   if (HAS_BIAS)
     bias_g = bias + (long)(bias_batch * heads + head) * Lq * Lk;
-  if (NEED_DBIAS)
-    dbias_g = dbias + (long)(bias_batch * heads + head) * Lq * Lk;
+  if (NEED_DBIAS) {
+    // fold into one of `dbias_chunks` scratch copies: the repeat group
+    // splits across chunks, cutting same-address atomic chain depth
+    const int chunk = (batch % bias_repeat) % dbias_chunks;
+    dbias_g = dbias + (long)chunk * dbias_stride
+        + (long)(bias_batch * heads + head) * Lq * Lk;
+  }
 
   const int kv_rows = min(BK, Lk - ktile * BK);
   stage_tile(k_g, k.rs, kv_rows, k_lds);
@@ -1063,8 +1069,10 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
     dv = at::empty({B, Lk, H, DH}, q.options()).permute({0, 2, 1, 3});
   }
   at::Tensor dbias;
+  int dbias_chunks = 1;
   if (need_dbias) {
-    dbias = at::zeros({B / bias_repeat, H, Lq, Lk},
+    dbias_chunks = (bias_repeat >= 8) ? 8 : 1;
+    dbias = at::zeros({dbias_chunks, B / bias_repeat, H, Lq, Lk},
                       q.options().dtype(at::kFloat));
   }
   const bool has_bias = bias.has_value();
@@ -1111,6 +1119,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
       has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
       dov, lse.data_ptr<float>(), delta.data_ptr<float>(), dkv, dvv,          \
       DB ? dbias.data_ptr<float>() : nullptr,                                 \
+      dbias_chunks, need_dbias ? dbias.stride(0) : 0L,                        \
       Lq, Lk, H, (int)bias_repeat, (float)scale)
 
   if (need_dbias) {
@@ -1123,6 +1132,6 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
 #undef DISPATCH_DKV
 
   std::vector<at::Tensor> ret = {dq, dk, dv};
-  if (need_dbias) ret.push_back(dbias);
+  if (need_dbias) ret.push_back(dbias.sum(0));
   return ret;
 }
