@@ -1069,9 +1069,11 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
     dv = at::empty({B, Lk, H, DH}, q.options()).permute({0, 2, 1, 3});
   }
   at::Tensor dbias;
+  // chunks > 1 shortens same-address atomic chains but the extra
+  // zeros+sum traffic measured NET-NEGATIVE (tri bwd 4.0 -> 4.3 ms at
+  // chunks=8, tools/attn_bench.py) — keep a single copy
   int dbias_chunks = 1;
   if (need_dbias) {
-    dbias_chunks = (bias_repeat >= 8) ? 8 : 1;
     dbias = at::zeros({dbias_chunks, B / bias_repeat, H, Lq, Lk},
                       q.options().dtype(at::kFloat));
   }
