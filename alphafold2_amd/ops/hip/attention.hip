@@ -426,27 +426,22 @@ void attn_fwd_kernel(TView q, TView k, TView v,
       }
     }
 
-    // online softmax; the O/l rescale is skipped exactly when the
-    // running max is unchanged (common after the first tiles)
-    float m_new[4];
+    // online softmax
+    float alpha[4], m_new[4];
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       float tile_max = rowred_max(s, reg);
       m_new[reg] = fmaxf(m_i[reg], tile_max);
+      alpha[reg] = (m_i[reg] <= NEG_INF) ? 0.f : __expf(m_i[reg] - m_new[reg]);
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
         s[c][reg] = (m_new[reg] <= NEG_INF) ? 0.f
             : __expf(s[c][reg] - m_new[reg]);
       }
-      if (m_new[reg] != m_i[reg]) {
-        const float alpha = (m_i[reg] <= NEG_INF)
-            ? 0.f : __expf(m_i[reg] - m_new[reg]);
-        l_i[reg] *= alpha;
+      l_i[reg] = l_i[reg] * alpha[reg] + rowred_sum(s, reg);
+      m_i[reg] = m_new[reg];
 #pragma unroll
-        for (int c = 0; c < 4; ++c) o_acc[c][reg] *= alpha;
-        m_i[reg] = m_new[reg];
-      }
-      l_i[reg] += rowred_sum(s, reg);
+      for (int c = 0; c < 4; ++c) o_acc[c][reg] *= alpha[reg];
     }
 
     // P tile -> per-wave swizzled LDS scratch (C layout -> A layout)
