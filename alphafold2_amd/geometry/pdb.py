@@ -11,7 +11,6 @@ import string
 import numpy as np
 import torch
 
-from .. import constants
 from ..vocab import VOCAB
 
 

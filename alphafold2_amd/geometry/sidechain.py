@@ -12,7 +12,7 @@ one vectorized step from per-residue idealized internal coordinates.
 import torch
 
 from .. import constants
-from ..vocab import VOCAB, CUSTOM_INFO, SC_BUILD, AA_LETTERS, PAD_CHAR
+from ..vocab import VOCAB, CUSTOM_INFO, SC_BUILD, PAD_CHAR
 from .backend import expand_dims_to
 
 # ---------------------------------------------------------------------------

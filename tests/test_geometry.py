@@ -188,3 +188,47 @@ def test_numpy_backend_agreement():
     t = RMSD(a, b)
     n = RMSD(a.numpy(), b.numpy())
     assert np.allclose(t.numpy(), n, atol=1e-5)
+
+
+def test_prot_covalent_bond_and_adjacency():
+    from alphafold2_amd.utils import prot_covalent_bond, nth_deg_adjacency
+    seqs = torch.tensor([[0, 5, 7]])  # A, G, I
+    mask_mat, attr_mat = prot_covalent_bond(seqs, adj_degree=1)
+    assert mask_mat.any()
+    # peptide bond between residue CA chain: symmetric adjacency
+    assert torch.equal(attr_mat[0], attr_mat[0].t())
+    adj = torch.zeros(5, 5)
+    adj[0, 1] = adj[1, 0] = adj[1, 2] = adj[2, 1] = 1
+    new_adj, attr = nth_deg_adjacency(adj, n=2)
+    assert attr[0, 2] == 2  # two hops
+
+
+def test_coords2pdb_writer(tmp_path):
+    from alphafold2_amd.utils import coords2pdb, scn_cloud_mask
+    seq = torch.tensor([0, 5, 3])  # A G D
+    cloud = scn_cloud_mask(seq[None])[0]
+    n_atoms = int(cloud.sum())
+    coords = torch.randn(n_atoms, 3)
+    path = coords2pdb(seq, coords, cloud, prefix=str(tmp_path) + '/')
+    text = open(path).read()
+    assert text.count('ATOM') == n_atoms
+    assert 'ALA' in text and 'GLY' in text and 'ASP' in text
+
+
+def test_mds_numpy_backend():
+    import numpy as np
+    from alphafold2_amd.utils import MDScaling
+    pts = np.random.randn(20, 3)
+    dist = np.linalg.norm(pts[:, None] - pts[None], axis=-1)
+    coords, _ = MDScaling(dist, iters=30, fix_mirror=False)
+    assert coords.shape == (1, 3, 20)
+
+
+def test_distmat_loss_options():
+    from alphafold2_amd.utils import distmat_loss_torch
+    a = torch.randn(10, 3)
+    b = torch.randn(10, 3)
+    l1 = distmat_loss_torch(a, b, p=2, q=1)         # MAE-style
+    l2 = distmat_loss_torch(a, b, clamp=(0, 5))
+    l3 = distmat_loss_torch(a, b, custom=lambda x, y: (x - y).abs())
+    assert all(torch.isfinite(t) for t in (l1, l2, l3))

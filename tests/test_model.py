@@ -332,3 +332,39 @@ def test_overfit_single_batch_cpu():
         opt.step()
         losses.append(float(loss))
     assert losses[-1] < losses[0] * 0.7, losses
+
+
+def test_seq_index_custom_numbering():
+    model = tiny_model().eval()
+    seq = torch.randint(0, 21, (1, 16))
+    msa = torch.randint(0, 21, (1, 2, 16))
+    with torch.no_grad():
+        r1 = model(seq, msa)
+        # chain-broken numbering (gap of 100 between residues 8 and 9)
+        idx = torch.cat([torch.arange(8), torch.arange(108, 116)])
+        r2 = model(seq, msa, seq_index=idx)
+    assert not torch.allclose(r1.distance, r2.distance)
+
+
+def test_disable_token_embed():
+    model = tiny_model(disable_token_embed=True).eval()
+    seq = torch.randint(0, 21, (1, 12))
+    msa = torch.randint(0, 21, (1, 2, 12))
+    seq_embed = torch.randn(1, 12, 32)
+    msa_embed = torch.randn(1, 2, 12, 32)
+    with torch.no_grad():
+        ret = model(seq, msa, seq_embed=seq_embed, msa_embed=msa_embed)
+    assert ret.distance.shape == (1, 12, 12, 37)
+    import pytest
+    with pytest.raises(AssertionError):
+        model(seq, msa)  # embeddings are mandatory in this mode
+
+
+def test_return_trunk_skips_structure():
+    model = tiny_model(predict_coords=True, structure_module_depth=1)
+    model.eval()
+    seq = torch.randint(0, 21, (1, 8))
+    msa = torch.randint(0, 21, (1, 2, 8))
+    with torch.no_grad():
+        ret = model(seq, msa, return_trunk=True)
+    assert ret.distance is not None  # trunk output, no coords

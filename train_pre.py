@@ -44,8 +44,7 @@ def main():
 
     from alphafold2_amd import Alphafold2
     from alphafold2_amd.data import SyntheticProteinDataset
-    from alphafold2_amd.parallel import (DataParallelEngine, all_reduce_mean,
-                                         init_distributed)
+    from alphafold2_amd.parallel import DataParallelEngine, init_distributed
     from alphafold2_amd.runtime import load_checkpoint, save_checkpoint
     from alphafold2_amd.utils import get_bucketed_distance_matrix
 
