@@ -73,7 +73,8 @@ def mds_numpy(pre_dist_mat, weights=None, iters=10, tol=1e-5, eigen=False,
         weights = np.ones_like(pre_dist_mat)
     pre_dist_mat = expand_dims_to(pre_dist_mat, length=3 - len(pre_dist_mat.shape))
     batch, N, _ = pre_dist_mat.shape
-    his = [np.inf]
+    # per-batch history (a scalar seed breaks numpy>=2 stacking)
+    his = [np.full(batch, np.inf)]
     best_stress = np.inf * np.ones(batch)
     best_3d_coords = 2 * np.random.rand(batch, 3, N) - 1
     for i in range(iters):

@@ -205,14 +205,14 @@ def test_prot_covalent_bond_and_adjacency():
 
 def test_coords2pdb_writer(tmp_path):
     from alphafold2_amd.utils import coords2pdb, scn_cloud_mask
-    seq = torch.tensor([0, 5, 3])  # A G D
+    seq = torch.tensor([0, 5, 3])  # A G E (alphabetical vocab)
     cloud = scn_cloud_mask(seq[None])[0]
     n_atoms = int(cloud.sum())
     coords = torch.randn(n_atoms, 3)
     path = coords2pdb(seq, coords, cloud, prefix=str(tmp_path) + '/')
     text = open(path).read()
     assert text.count('ATOM') == n_atoms
-    assert 'ALA' in text and 'GLY' in text and 'ASP' in text
+    assert 'ALA' in text and 'GLY' in text and 'GLU' in text
 
 
 def test_mds_numpy_backend():
