@@ -368,3 +368,26 @@ def test_return_trunk_skips_structure():
     with torch.no_grad():
         ret = model(seq, msa, return_trunk=True)
     assert ret.distance is not None  # trunk output, no coords
+
+
+def test_rotary_embeddings_functional():
+    from alphafold2_amd.models.rotary import (
+        AxialRotaryEmbedding, DepthWiseConv1d, FixedPositionalEmbedding,
+        apply_rotary_pos_emb)
+    fp = FixedPositionalEmbedding(dim=16)
+    sin, cos = fp(10, torch.device('cpu'))
+    assert sin.shape == (10, 16)
+    q = torch.randn(2, 10, 16)
+    k = torch.randn(2, 10, 16)
+    q2, k2 = apply_rotary_pos_emb(q, k, (sin, cos))
+    assert q2.shape == q.shape
+    # rotary preserves norms per position
+    assert torch.allclose(q2.norm(dim=-1), q.norm(dim=-1), atol=1e-5)
+
+    ax = AxialRotaryEmbedding(dim=16)
+    sin2, cos2 = ax(6, torch.device('cpu'))
+    assert sin2.shape[0] == 36
+
+    conv = DepthWiseConv1d(8, 16, kernel_size=3, padding=1)
+    y = conv(torch.randn(2, 8, 12))
+    assert y.shape == (2, 16, 12)
