@@ -1,1 +1,2 @@
 from .synthetic import SyntheticProteinDataset, synthetic_batch
+from .trrosetta import TrRosettaDataset, TrRosettaDataModule, collate_batch

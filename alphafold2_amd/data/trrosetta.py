@@ -135,3 +135,57 @@ def collate_batch(items, pad_id=20):
     if has_coords:
         out['coords'] = coords
     return out
+
+
+class TrRosettaDataModule:
+    """Train/val/test split + loader factory — capability parity with the
+    reference's Lightning ``TrRosettaDataModule``
+    (training_scripts/datasets/trrosetta.py:352-476), re-designed for the
+    MI355X training stack: no Lightning dependency, and the loaders are
+    DistributedSampler-aware so the same module drives one-process-per-GPU
+    DP over RCCL (each rank sees a disjoint shard).
+
+    Splits are deterministic (seeded permutation of the id list), so every
+    rank computes identical splits without communication.
+    """
+
+    def __init__(self, root, train_frac=0.9, val_frac=0.05, batch_size=1,
+                 max_seq_len=250, crop_len=256, max_msa_depth=32,
+                 num_workers=0, seed=0, **dataset_kwargs):
+        self.batch_size = batch_size
+        self.num_workers = num_workers
+        base = TrRosettaDataset(root, max_seq_len=max_seq_len,
+                                crop_len=crop_len,
+                                max_msa_depth=max_msa_depth,
+                                seed=seed, **dataset_kwargs)
+        n = len(base)
+        g = torch.Generator().manual_seed(seed)
+        perm = torch.randperm(n, generator=g).tolist()
+        n_train = max(1, int(n * train_frac))
+        n_val = int(n * val_frac)
+        self.train_set = torch.utils.data.Subset(base, perm[:n_train])
+        self.val_set = torch.utils.data.Subset(
+            base, perm[n_train:n_train + n_val] or perm[:1])
+        self.test_set = torch.utils.data.Subset(
+            base, perm[n_train + n_val:] or perm[:1])
+
+    def _loader(self, ds, shuffle):
+        import torch.distributed as dist
+        sampler = None
+        if dist.is_available() and dist.is_initialized() \
+                and dist.get_world_size() > 1:
+            sampler = torch.utils.data.distributed.DistributedSampler(
+                ds, shuffle=shuffle)
+            shuffle = False
+        return torch.utils.data.DataLoader(
+            ds, batch_size=self.batch_size, shuffle=shuffle, sampler=sampler,
+            num_workers=self.num_workers, collate_fn=collate_batch)
+
+    def train_dataloader(self):
+        return self._loader(self.train_set, shuffle=True)
+
+    def val_dataloader(self):
+        return self._loader(self.val_set, shuffle=False)
+
+    def test_dataloader(self):
+        return self._loader(self.test_set, shuffle=False)
