@@ -71,18 +71,18 @@ def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
     dropout (reference alphafold2.py:172) routes to the eager path."""
     drop_active = dropout > 0. and training
     fusable = (
-        tie_dim is None
-        and not drop_active
+        not drop_active
         and q.dtype == torch.bfloat16
         and q.shape[-1] == 64
         and (bias is None or bias.dtype == torch.bfloat16)
+        and (tie_dim is None or q.shape[0] % tie_dim == 0)
         and using_hip(q, 'attn_fwd')
     )
     if fusable:
         from .hip_autograd import hip_attention_core
         return hip_attention_core(q, k, v, bias=bias, mask=mask,
                                   context_mask=context_mask,
-                                  bias_repeat=bias_repeat)
+                                  tie_dim=tie_dim, bias_repeat=bias_repeat)
     if bias is not None and bias_repeat != 1:
         bias = bias.repeat_interleave(bias_repeat, dim=0)
     return eager.attention_core(q, k, v, bias=bias, mask=mask,

@@ -85,12 +85,67 @@ class _AttentionFn(torch.autograd.Function):
 
 def hip_attention_core(q, k, v, bias=None, mask=None, context_mask=None,
                        tie_dim=None, bias_repeat=1):
-    assert tie_dim is None, "tie_dim handled by the eager path"
     key_mask = context_mask if context_mask is not None else mask
     if key_mask is not None:
         key_mask = key_mask.to(torch.uint8)
     scale = q.shape[-1] ** -0.5
+    if tie_dim is not None:
+        return _AttentionTiedFn.apply(q, k, v, bias, key_mask, tie_dim,
+                                      bias_repeat, scale)
     return _AttentionFn.apply(q, k, v, bias, key_mask, bias_repeat, scale)
+
+
+class _AttentionTiedFn(torch.autograd.Function):
+    """Tied-query ("global column", reference alphafold2.py:142-151)
+    attention on the fused kernels: queries are MEANED over groups of
+    `tie_dim` consecutive batch entries and the group-mean query attends
+    to each entry's own keys/values.  K2 in SURVEY.md §2.17.
+
+    Runs as standard attention with the group-mean query replicated
+    across the group (one bf16 copy); backward group-sums dq_exp and
+    divides by tie_dim (the mean+broadcast chain rule), so dq matches
+    plain autograd exactly.
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias, mask, tie_dim, bias_repeat, scale):
+        ext = _load_ext()
+        Bh = q.shape[0]
+        b = Bh // tie_dim
+        qm = q.reshape(b, tie_dim, *q.shape[1:]).mean(dim=1, keepdim=True)
+        q_exp = qm.expand(b, tie_dim, *qm.shape[2:]) \
+                  .reshape(Bh, *qm.shape[2:]).contiguous()
+        bias_c = bias.contiguous() if bias is not None else None
+        mask_c = mask.contiguous() if mask is not None else None
+        out, lse = ext.attn_fwd(q_exp, k, v, bias_c, mask_c, bias_repeat,
+                                scale)
+        ctx.save_for_backward(q_exp, k, v, out, lse,
+                              *([bias_c] if bias_c is not None else []))
+        ctx.has_bias = bias_c is not None
+        ctx.mask = mask_c
+        ctx.meta = (tie_dim, bias_repeat, scale)
+        ctx.bias_requires_grad = bias is not None and bias.requires_grad
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _load_ext()
+        saved = ctx.saved_tensors
+        q_exp, k, v, out, lse = saved[:5]
+        bias = saved[5] if ctx.has_bias else None
+        tie_dim, bias_repeat, scale = ctx.meta
+        need_dbias = ctx.bias_requires_grad
+        rets = ext.attn_bwd(dout, q_exp, k, v, out, lse, bias,
+                            ctx.mask, bias_repeat, scale, need_dbias)
+        dq_exp, dk, dv = rets[:3]
+        Bh = dq_exp.shape[0]
+        b = Bh // tie_dim
+        dq = dq_exp.reshape(b, tie_dim, *dq_exp.shape[1:]) \
+                   .sum(dim=1, keepdim=True).div_(tie_dim) \
+                   .expand(b, tie_dim, *dq_exp.shape[1:]) \
+                   .reshape(Bh, *dq_exp.shape[1:])
+        dbias = rets[3].to(bias.dtype) if need_dbias else None
+        return dq, dk, dv, dbias, None, None, None, None
 
 
 class _AttentionPackedFn(torch.autograd.Function):
