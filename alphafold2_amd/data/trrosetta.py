@@ -34,7 +34,7 @@ def encode_seq(seq: str):
 class TrRosettaDataset(torch.utils.data.Dataset):
     def __init__(self, root, max_seq_len=250, crop_len=256, max_msa_depth=32,
                  min_msa_depth=1, num_buckets=constants.DISTOGRAM_BUCKETS,
-                 seed=0):
+                 seed=0, cache_dir=None):
         self.root = root
         self.max_seq_len = max_seq_len
         self.crop_len = crop_len
@@ -42,6 +42,11 @@ class TrRosettaDataset(torch.utils.data.Dataset):
         self.min_msa_depth = min_msa_depth
         self.num_buckets = num_buckets
         self.rng = random.Random(seed)
+        # per-item parse cache (reference trrosetta.py:178-200 parity):
+        # a3m parsing dominates cold reads; cache the tokenized arrays
+        self.cache_dir = cache_dir
+        if cache_dir is not None:
+            os.makedirs(cache_dir, exist_ok=True)
 
         ids = set()
         for fn in os.listdir(root):
@@ -56,6 +61,29 @@ class TrRosettaDataset(torch.utils.data.Dataset):
         return len(self.ids)
 
     def _load(self, pid):
+        if self.cache_dir is not None:
+            cpath = os.path.join(self.cache_dir, pid + '.cache.npz')
+            if os.path.exists(cpath):
+                data = np.load(cpath, allow_pickle=False)
+                return (torch.as_tensor(data['msa']).long(),
+                        torch.as_tensor(data['xyz']).float()
+                        if 'xyz' in data else None,
+                        torch.as_tensor(data['dist']).float()
+                        if 'dist' in data else None)
+        msa, coords, dist = self._load_raw(pid)
+        if self.cache_dir is not None:
+            arrays = {'msa': msa.numpy()}
+            if coords is not None:
+                arrays['xyz'] = coords.numpy()
+            if dist is not None:
+                arrays['dist'] = dist.numpy()
+            tmp = cpath + '.tmp'
+            np.savez(tmp, **arrays)
+            # np.savez appends .npz to paths without the suffix
+            os.replace(tmp if os.path.exists(tmp) else tmp + '.npz', cpath)
+        return msa, coords, dist
+
+    def _load_raw(self, pid):
         npz_path = os.path.join(self.root, pid + '.npz')
         a3m_path = os.path.join(self.root, pid + '.a3m')
         msa = coords = dist = None

@@ -96,6 +96,20 @@ def test_trrosetta_dataset(tmp_path):
     assert ret.distance.shape == (2, n, n, 37)
 
 
+def test_trrosetta_cache(tmp_path):
+    from alphafold2_amd.data.trrosetta import TrRosettaDataset
+    with open(tmp_path / 'p.a3m', 'w') as f:
+        f.write('>q\nACDEFGHIKLMNPQRSTVWY\n>h\nACDEFGHIKLMNPQRSTVWY\n')
+    cache = str(tmp_path / 'cache')
+    ds = TrRosettaDataset(str(tmp_path), crop_len=32, max_msa_depth=4,
+                          cache_dir=cache)
+    it1 = ds[0]
+    assert os.path.exists(os.path.join(cache, 'p.cache.npz'))
+    it2 = ds[0]  # served from cache
+    assert torch.equal(it1['seq'], it2['seq'])
+    assert torch.equal(it1['msa'], it2['msa'])
+
+
 def test_trrosetta_datamodule(tmp_path):
     from alphafold2_amd.data import TrRosettaDataModule
     for i in range(6):
