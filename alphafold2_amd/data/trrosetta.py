@@ -217,3 +217,39 @@ class TrRosettaDataModule:
 
     def test_dataloader(self):
         return self._loader(self.test_set, shuffle=False)
+
+
+def add_argparse_args(parser):
+    """Reference-parity argparse config for the data module
+    (training_scripts/datasets/trrosetta.py:353-373)."""
+    g = parser.add_argument_group('TrRosettaDataModule')
+    g.add_argument('--data-root', type=str, default=None)
+    g.add_argument('--data-cache', type=str, default=None)
+    g.add_argument('--train-frac', type=float, default=0.9)
+    g.add_argument('--val-frac', type=float, default=0.05)
+    g.add_argument('--data-batch-size', type=int, default=1)
+    g.add_argument('--data-max-seq-len', type=int, default=250)
+    g.add_argument('--data-crop-len', type=int, default=256)
+    g.add_argument('--data-msa-depth', type=int, default=32)
+    g.add_argument('--data-workers', type=int, default=0)
+    return parser
+
+
+def from_argparse_args(args):
+    return TrRosettaDataModule(
+        args.data_root, train_frac=args.train_frac, val_frac=args.val_frac,
+        batch_size=args.data_batch_size, max_seq_len=args.data_max_seq_len,
+        crop_len=args.data_crop_len, max_msa_depth=args.data_msa_depth,
+        num_workers=args.data_workers, cache_dir=args.data_cache)
+
+
+if __name__ == '__main__':
+    # dataset smoke test (reference trrosetta.py:479-497 parity):
+    # point at a local directory and print one collated batch's shapes
+    import argparse
+    ap = add_argparse_args(argparse.ArgumentParser())
+    cli = ap.parse_args()
+    dm = from_argparse_args(cli)
+    batch = next(iter(dm.train_dataloader()))
+    for k_, v_ in batch.items():
+        print(k_, tuple(v_.shape), v_.dtype)
