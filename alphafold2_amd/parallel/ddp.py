@@ -70,7 +70,8 @@ def all_reduce_mean(t: torch.Tensor):
 
 
 class _Bucket:
-    __slots__ = ('params', 'flat', 'numel', 'ready', 'work', 'offsets')
+    __slots__ = ('params', 'flat', 'numel', 'ready', 'work', 'offsets',
+                 'seen')
 
     def __init__(self):
         self.params: List[torch.nn.Parameter] = []
@@ -79,6 +80,7 @@ class _Bucket:
         self.ready = 0
         self.work = None
         self.offsets = {}
+        self.seen = set()
 
 
 class DataParallelEngine:
@@ -170,6 +172,7 @@ class DataParallelEngine:
         self._ensure_flat(bucket, p.grad)
         off = bucket.offsets[p]
         bucket.flat[off:off + p.numel()].copy_(p.grad.reshape(-1))
+        bucket.seen.add(p)
         bucket.ready += 1
         if bucket.ready == len(bucket.params):
             # one large message per bucket; async so backward keeps going
@@ -198,8 +201,15 @@ class DataParallelEngine:
                     p.grad.copy_(
                         bucket.flat[off:off + p.numel()].view_as(p.grad))
             bucket.ready = 0
+            bucket.seen.clear()
 
     def _flush_partial(self, bucket: _Bucket):
+        """Reduce a bucket some of whose params never fired their hook
+        this step.  A param can hold a REAL accumulated grad and still
+        not fire (it got its grad during a no_sync micro-batch and was
+        unused in the final one), so copy every un-seen grad into the
+        flat buffer — and zero the slice of truly grad-less params so a
+        previous step's values never leak into the collective."""
         device = next(self.model.parameters()).device
         any_grad = next((p.grad for p in bucket.params if p.grad is not None),
                         None)
@@ -208,6 +218,15 @@ class DataParallelEngine:
                 bucket, torch.zeros(1, device=device))
         elif bucket.flat is None:
             self._ensure_flat(bucket, any_grad)
+        for p in bucket.params:
+            if p in bucket.seen:
+                continue
+            off = bucket.offsets[p]
+            dst = bucket.flat[off:off + p.numel()]
+            if p.grad is not None:
+                dst.copy_(p.grad.reshape(-1))
+            else:
+                dst.zero_()
         bucket.work = dist.all_reduce(bucket.flat, op=dist.ReduceOp.SUM,
                                       group=self.group, async_op=True)
         bucket.ready = len(bucket.params)
@@ -217,6 +236,7 @@ class DataParallelEngine:
         for bucket in self._buckets:
             bucket.ready = 0
             bucket.work = None
+            bucket.seen.clear()
 
     def remove(self):
         for h in self._hooks:

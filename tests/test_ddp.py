@@ -120,3 +120,73 @@ def test_ddp_no_sync_accumulation():
         p.join(timeout=60)
         assert p.exitcode == 0
     assert torch.allclose(results[0], results[1], atol=1e-6)
+
+
+class _TwoPath(torch.nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.a = torch.nn.Linear(8, 8)
+        self.b = torch.nn.Linear(8, 8)
+
+    def forward(self, x, use_b=True):
+        out = self.a(x)
+        if use_b:
+            out = out + self.b(x)
+        return out
+
+
+def _unused_param_worker(rank, port, q):
+    """A param that gets its grad in a no_sync micro-batch but is NOT
+    used in the final (sync) micro-batch must still be reduced."""
+    from alphafold2_amd.parallel import DataParallelEngine
+    _setup(rank, WORLD, port)
+    torch.manual_seed(5)
+    model = _TwoPath()
+    engine = DataParallelEngine(model, bucket_cap_mb=64)
+
+    torch.manual_seed(rank * 31)
+    x1 = torch.randn(2, 8)
+    x2 = torch.randn(2, 8)
+    with engine.no_sync():
+        model(x1, use_b=True).pow(2).sum().backward()
+    model(x2, use_b=False).pow(2).sum().backward()  # b unused here
+    engine.finalize()
+    grads = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
+    q.put((rank, x1, x2, grads))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_ddp_unused_param_in_sync_micro():
+    ctx = mp.get_context('spawn')
+    q = ctx.SimpleQueue()
+    port = 29515
+    procs = [ctx.Process(target=_unused_param_worker, args=(r, port, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, x1, x2, grads = q.get()
+        results[rank] = (x1, x2, grads)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    assert torch.allclose(results[0][2], results[1][2], atol=1e-6)
+
+    # equals the all-rank average of locally accumulated gradients
+    torch.manual_seed(5)
+    model = _TwoPath()
+    expected = None
+    for rank in range(WORLD):
+        model.zero_grad()
+        x1, x2 = results[rank][0], results[rank][1]
+        model(x1, use_b=True).pow(2).sum().backward()
+        model(x2, use_b=False).pow(2).sum().backward()
+        g = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
+        expected = g if expected is None else expected + g
+    expected = expected / WORLD
+    assert torch.allclose(results[0][2], expected, atol=1e-5), \
+        (results[0][2] - expected).abs().max()
