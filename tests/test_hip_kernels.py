@@ -150,7 +150,10 @@ def _eager_ref(q, k, v, bias=None, mask=None, bias_repeat=1):
 
 
 @pytest.mark.parametrize("B,h,Lq,Lk", [(4, 8, 128, 128), (2, 2, 256, 256),
-                                       (3, 2, 100, 72), (2, 8, 64, 257)])
+                                       (3, 2, 100, 72), (2, 8, 64, 257),
+                                       # template-pointwise shape (Lq=1,
+                                       # tiny Lk) and sub-tile sizes
+                                       (8, 2, 1, 10), (2, 2, 7, 3)])
 @pytest.mark.parametrize("use_bias", [False, True])
 @pytest.mark.parametrize("use_mask", [False, True])
 def test_attn_fwd_bwd_parity(ext, B, h, Lq, Lk, use_bias, use_mask):
