@@ -457,7 +457,9 @@ void attn_fwd_kernel(TView q, TView k, TView v,
       }
     }
     // wave-internal LDS dependency: ds ops from this wave only
-    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0)
+    // LDS-only wait: vmcnt stays open so the prefetched next
+    // K/V tile's global loads keep flying under the PV MFMAs
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     bf16x8 p_frag[2];
 #pragma unroll
@@ -677,7 +679,7 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
             = (bf16_t)ds;
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     bf16x8 ds_frag[2];
 #pragma unroll
@@ -874,7 +876,7 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
             = (bf16_t)pt[c][reg];
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     bf16x8 pt_frag[2];
 #pragma unroll
     for (int kblk = 0; kblk < 2; ++kblk)
@@ -922,7 +924,7 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
             = (bf16_t)ds_total;
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     bf16x8 dst_frag[2];
 #pragma unroll
     for (int kblk = 0; kblk < 2; ++kblk)
