@@ -241,6 +241,51 @@ __device__ __forceinline__ void stage_store_n(const StageRegsN<NT, R>& r,
   }
 }
 
+// column-wise-mapped variant for tiles that are ONLY stored transposed
+// (fwd V^T): thread idx -> (row = idx&63, chunk = idx>>6), so within a
+// wave the transposed-store byte offset `row*2` spans the full 0..126
+// range -> ~2-way LDS bank conflict instead of the 16-way conflict of
+// the row-wise mapping (where row*2 spans only 0..14 for fixed j).
+// Global-load cost: each 128-byte row cacheline is read in 16B chunks
+// by 8 different waves (L2-served) instead of one coalesced pull.
+template <int NT>
+__device__ __forceinline__ void stage_load_colwise_n(
+    const bf16_t* __restrict__ g, long row_stride, int rows,
+    StageRegsN<NT, 64>& r) {
+  constexpr int P = (64 * 8 + NT - 1) / NT;
+#pragma unroll
+  for (int pass = 0; pass < P; ++pass) {
+    int idx = threadIdx.x + pass * NT;
+    int row = idx & 63;
+    int c16 = (idx >> 6) << 4;
+    float4 val = {0, 0, 0, 0};
+    if (row < rows) {
+      val = *reinterpret_cast<const float4*>(
+          reinterpret_cast<const char*>(g + row * row_stride) + c16);
+    }
+    r.v[pass] = val;
+  }
+}
+
+template <int NT>
+__device__ __forceinline__ void stage_store_t_colwise_n(
+    const StageRegsN<NT, 64>& r, char* lds) {
+  constexpr int P = (64 * 8 + NT - 1) / NT;
+#pragma unroll
+  for (int pass = 0; pass < P; ++pass) {
+    int idx = threadIdx.x + pass * NT;
+    int row = idx & 63;
+    int c8 = (idx >> 6) << 3;
+    const float4 val = r.v[pass];
+    const bf16_t* vv = reinterpret_cast<const bf16_t*>(&val);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      *reinterpret_cast<bf16_t*>(
+          lds + swz(c8 + j, row * (int)sizeof(bf16_t))) = vv[j];
+    }
+  }
+}
+
 template <int NT, int R>
 __device__ __forceinline__ void stage_store_t_n(const StageRegsN<NT, R>& r,
                                                 char* lds) {
@@ -365,12 +410,12 @@ void attn_fwd_kernel(TView q, TView k, TView v,
   const int n_kv = (Lk + BK - 1) / BK;
   StageRegsN<FNT, BK> kreg, vreg;
   stage_load_n<FNT, BK>(k_g, k.rs, min(BK, Lk), kreg);
-  stage_load_n<FNT, BK>(v_g, v.rs, min(BK, Lk), vreg);
+  stage_load_colwise_n<FNT>(v_g, v.rs, min(BK, Lk), vreg);
   for (int t = 0; t < n_kv; ++t) {
     const int kv_rows = min(BK, Lk - t * BK);
     __syncthreads();
     stage_store_n<FNT, BK>(kreg, k_lds);
-    stage_store_t_n<FNT, BK>(vreg, vt_lds);
+    stage_store_t_colwise_n<FNT>(vreg, vt_lds);
     if (HAS_MASK && threadIdx.x < BK) {
       m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
           ? mask[(long)batch * Lk + t * BK + threadIdx.x] : 0;
@@ -381,8 +426,8 @@ void attn_fwd_kernel(TView q, TView k, TView v,
       const int next_rows = min(BK, Lk - (t + 1) * BK);
       stage_load_n<FNT, BK>(k_g + (long)(t + 1) * BK * k.rs, k.rs,
                             next_rows, kreg);
-      stage_load_n<FNT, BK>(v_g + (long)(t + 1) * BK * v.rs, v.rs,
-                            next_rows, vreg);
+      stage_load_colwise_n<FNT>(v_g + (long)(t + 1) * BK * v.rs, v.rs,
+                                next_rows, vreg);
     }
 
     // S = Q K^T  (16 q x 64 kv per wave); setprio favors the MFMA
