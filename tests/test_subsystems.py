@@ -272,3 +272,25 @@ def test_predict_cli_a3m(tmp_path):
         cwd=root, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-2000:]
     assert os.path.exists(out_pdb)
+
+
+def test_profiling_helpers(tmp_path):
+    from alphafold2_amd.runtime.profiling import (
+        StepTimer, kernel_stats_summary, profile_trace)
+    t = StepTimer(sync_cuda=False)
+    for _ in range(5):
+        with t:
+            torch.randn(64, 64) @ torch.randn(64, 64)
+    s = t.summary()
+    assert s['steps'] == 5 and s['p50_ms'] <= s['max_ms']
+
+    # parse the committed round-1 rocprofv3 evidence
+    rows = kernel_stats_summary('profiles/r01_final_kernel_stats.csv', top=5)
+    assert len(rows) == 5
+    assert rows[0]['total_ms'] >= rows[1]['total_ms']
+    assert all(0 <= r['pct'] <= 100 for r in rows)
+
+    trace = str(tmp_path / 'trace.json')
+    with profile_trace(trace):
+        torch.randn(8, 8) @ torch.randn(8, 8)
+    assert os.path.getsize(trace) > 0
