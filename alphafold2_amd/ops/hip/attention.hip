@@ -353,7 +353,9 @@ constexpr int FBQ = 128;    // fwd query rows per workgroup
 constexpr int FNT = 512;    // fwd threads (8 waves x 16 q rows)
 
 template <bool HAS_BIAS, bool HAS_MASK>
-__global__ __launch_bounds__(FNT, 2)
+// minWavesPerEU=4 caps VGPRs at 128: the <bias,mask> variant was
+// 129 VGPRs = 3 waves/SIMD = only ONE 8-wave block resident per CU
+__global__ __launch_bounds__(FNT, 4)
 void attn_fwd_kernel(TView q, TView k, TView v,
                      const bf16_t* __restrict__ bias,
                      const unsigned char* __restrict__ mask,
@@ -361,10 +363,15 @@ void attn_fwd_kernel(TView q, TView k, TView v,
                      int Lq, int Lk, int heads, int bias_repeat,
                      float scale) {
   __shared__ char q_lds[FBQ * ROWB];
-  __shared__ char k_lds[BK * ROWB];
-  __shared__ char vt_lds[BK * ROWB];  // V transposed: [dv][kv]
+  // K / V^T / mask tiles are DOUBLE-buffered so the loop needs a single
+  // __syncthreads per KV tile (placed between store and compute): the
+  // barrier bounds the wave spread to one iteration, so store(t+1)
+  // writes buf[(t+1)&1] while the slowest wave still reads buf[t&1] —
+  // never the same buffer.  (~64KB LDS total: 2 blocks/CU preserved.)
+  __shared__ char k_lds[2][BK * ROWB];
+  __shared__ char vt_lds[2][BK * ROWB];  // V transposed: [dv][kv]
   __shared__ char p_lds[FNT / 64][16 * ROWB];
-  __shared__ unsigned char m_lds[BK];
+  __shared__ unsigned char m_lds[2][BK];
 
   const int qtile = blockIdx.x;
   const int bh = blockIdx.y;            // batch*heads + head
@@ -413,22 +420,23 @@ void attn_fwd_kernel(TView q, TView k, TView v,
   stage_load_colwise_n<FNT>(v_g, v.rs, min(BK, Lk), vreg);
   for (int t = 0; t < n_kv; ++t) {
     const int kv_rows = min(BK, Lk - t * BK);
-    __syncthreads();
-    stage_store_n<FNT, BK>(kreg, k_lds);
-    stage_store_t_colwise_n<FNT>(vreg, vt_lds);
+    const int buf = t & 1;
+    stage_store_n<FNT, BK>(kreg, k_lds[buf]);
+    stage_store_t_colwise_n<FNT>(vreg, vt_lds[buf]);
     if (HAS_MASK && threadIdx.x < BK) {
-      m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
+      m_lds[buf][threadIdx.x] = (threadIdx.x < kv_rows)
           ? mask[(long)batch * Lk + t * BK + threadIdx.x] : 0;
     }
-    __syncthreads();
     if (t + 1 < n_kv) {
-      // next tile's loads fly while this tile computes
+      // next tile's loads fly across the barrier and under compute
+      // (the ds_writes above read kreg/vreg at issue, in order)
       const int next_rows = min(BK, Lk - (t + 1) * BK);
       stage_load_n<FNT, BK>(k_g + (long)(t + 1) * BK * k.rs, k.rs,
                             next_rows, kreg);
       stage_load_colwise_n<FNT>(v_g + (long)(t + 1) * BK * v.rs, v.rs,
                                 next_rows, vreg);
     }
+    __syncthreads();
 
     // S = Q K^T  (16 q x 64 kv per wave); setprio favors the MFMA
     // cluster when co-resident waves are staging (guide T5)
@@ -439,7 +447,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
       f32x4 acc = {0, 0, 0, 0};
 #pragma unroll
       for (int dblk = 0; dblk < 2; ++dblk) {
-        bf16x8 kf = frag_row(k_lds, c * 16 + (lane & 15), dblk);
+        bf16x8 kf = frag_row(k_lds[buf], c * 16 + (lane & 15), dblk);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[dblk], kf, acc,
                                                       0, 0, 0);
       }
@@ -461,7 +469,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
     for (int c = 0; c < 4; ++c) {
       const int col = c * 16 + (lane & 15);
       const bool col_ok = col < kv_rows &&
-          (!HAS_MASK || m_lds[col]);
+          (!HAS_MASK || m_lds[buf][col]);
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         float val = s[c][reg] * scale;
@@ -518,7 +526,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
       f32x4 acc = o_acc[c];
 #pragma unroll
       for (int kblk = 0; kblk < 2; ++kblk) {
-        bf16x8 vf = frag_row(vt_lds, c * 16 + (lane & 15), kblk);
+        bf16x8 vf = frag_row(vt_lds[buf], c * 16 + (lane & 15), kblk);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[kblk], vf, acc,
                                                       0, 0, 0);
       }
@@ -577,7 +585,9 @@ __global__ void attn_delta_kernel(TView dout, TView out,
 // backward dQ: loop kv tiles; dS = P*(dP - delta)*scale; dQ += dS K
 
 template <bool HAS_BIAS, bool HAS_MASK>
-__global__ __launch_bounds__(256, 2)
+// minWavesPerEU=3 caps VGPRs at 170 (the <bias,mask> variant was 178
+// -> 2 waves/SIMD); 3 waves = 3 resident 4-wave blocks per CU
+__global__ __launch_bounds__(256, 3)
 void attn_bwd_dq_kernel(TView q, TView k, TView v,
                         const bf16_t* __restrict__ bias,
                         const unsigned char* __restrict__ mask,
