@@ -9,7 +9,7 @@ import numpy as np
 import torch
 
 from .backend import (
-    exists, set_backend_kwarg, expand_arg_dims, invoke_torch_or_numpy,
+    set_backend_kwarg, expand_arg_dims, invoke_torch_or_numpy,
 )
 
 # ---------------------------------------------------------------------------

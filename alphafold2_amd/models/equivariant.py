@@ -21,7 +21,6 @@ updated (single_repr, coords), and plug into Alphafold2's refinement
 loop behind `structure_module_type`.
 """
 import torch
-import torch.nn.functional as F
 from torch import nn
 
 

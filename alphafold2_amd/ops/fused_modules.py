@@ -1,7 +1,6 @@
 """nn.Module shims that keep standard parameter layouts (state-dict
 compatible with nn.LayerNorm etc.) while routing compute through the
 fused gfx950 kernels."""
-import torch
 from torch import nn
 
 from . import dispatch

@@ -7,7 +7,7 @@ import pytest
 import torch
 
 from alphafold2_amd import Alphafold2
-from alphafold2_amd.data import SyntheticProteinDataset, synthetic_batch
+from alphafold2_amd.data import SyntheticProteinDataset
 from alphafold2_amd.mlm import MLM
 from alphafold2_amd.runtime import load_checkpoint, save_checkpoint
 
