@@ -88,8 +88,11 @@ def prot_covalent_bond(seqs, adj_degree=1, cloud_mask=None, mat=True,
                 bonds = bonds[:, :-1]
             adj_mat[s, bonds[0], bonds[1]] = 1
         adj_mat[s] = adj_mat[s] + adj_mat[s].t()
-        adj_mat, attr_mat = nth_deg_adjacency(adj_mat, n=adj_degree,
-                                              sparse=sparse)
+    # power the adjacency ONCE over the whole batch (applying it inside
+    # the per-item loop would re-power earlier items' adjacency at every
+    # later item — wrong hop attributes for batch > 1, adj_degree >= 2)
+    adj_mat, attr_mat = nth_deg_adjacency(adj_mat, n=adj_degree,
+                                          sparse=sparse)
     if mat:
         return attr_mat.bool().to(device), attr_mat.to(device)
     edge_idxs = attr_mat[0].nonzero().t().long()

@@ -232,3 +232,14 @@ def test_distmat_loss_options():
     l2 = distmat_loss_torch(a, b, clamp=(0, 5))
     l3 = distmat_loss_torch(a, b, custom=lambda x, y: (x - y).abs())
     assert all(torch.isfinite(t) for t in (l1, l2, l3))
+
+
+def test_prot_covalent_bond_batch_independence():
+    """Each batch item's hop attributes must equal its single-item
+    computation (regression: nth_deg applied once, not per item)."""
+    from alphafold2_amd.utils import prot_covalent_bond
+    seqs = torch.tensor([[0, 5, 7], [3, 3, 3]])
+    _, attr_batch = prot_covalent_bond(seqs, adj_degree=2)
+    for i in range(2):
+        _, attr_single = prot_covalent_bond(seqs[i:i + 1], adj_degree=2)
+        assert torch.equal(attr_batch[i], attr_single[0])
