@@ -243,3 +243,61 @@ def test_prot_covalent_bond_batch_independence():
     for i in range(2):
         _, attr_single = prot_covalent_bond(seqs[i:i + 1], adj_degree=2)
         assert torch.equal(attr_batch[i], attr_single[0])
+
+
+# ---------------------------------------------------------------------------
+# property-based invariants (hypothesis)
+
+try:
+    from hypothesis import given, settings, strategies as st
+    HAS_HYP = True
+except ImportError:  # pragma: no cover
+    HAS_HYP = False
+
+if HAS_HYP:
+    @settings(max_examples=25, deadline=None)
+    @given(st.integers(min_value=4, max_value=64), st.integers(0, 10**6))
+    def test_kabsch_rigid_motion_invariant(n, seed):
+        """Kabsch must recover ANY rigid motion: RMSD(after) ~ 0."""
+        from alphafold2_amd.models.quaternion import quaternion_to_matrix
+        g = torch.Generator().manual_seed(seed)
+        a = torch.randn(3, n, generator=g).double()
+        q = torch.randn(4, generator=g).double()
+        q = q / q.norm()
+        R = quaternion_to_matrix(q[None])[0]
+        t = torch.randn(3, 1, generator=g).double() * 10
+        b = R @ a + t
+        a_, b_ = Kabsch(a, b)
+        assert RMSD(a_, b_).item() < 1e-6
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.integers(min_value=5, max_value=40), st.integers(0, 10**6))
+    def test_tmscore_bounds_and_self(n, seed):
+        g = torch.Generator().manual_seed(seed)
+        a = torch.randn(1, 3, n, generator=g)
+        b = torch.randn(1, 3, n, generator=g)
+        tm = TMscore(a, b)
+        assert 0.0 <= tm.item() <= 1.0
+        assert torch.allclose(TMscore(a, a), torch.ones(1))
+
+    @settings(max_examples=15, deadline=None)
+    @given(st.integers(min_value=6, max_value=24), st.integers(0, 10**6))
+    def test_lddt_bounds(n, seed):
+        g = torch.Generator().manual_seed(seed)
+        a = torch.randn(1, n, 14, 3, generator=g) * 5
+        b = a + torch.randn(1, n, 14, 3, generator=g) * 0.1
+        cloud = torch.ones(1, n, 14).bool()
+        val = lddt_ca_torch(a, b, cloud)
+        assert (val >= 0).all() and (val <= 1).all()
+
+    @settings(max_examples=15, deadline=None)
+    @given(st.integers(min_value=2, max_value=8), st.integers(0, 10**6))
+    def test_distogram_bucket_bounds(b, seed):
+        from alphafold2_amd.utils import get_bucketed_distance_matrix
+        g = torch.Generator().manual_seed(seed)
+        coords = torch.randn(b, 12, 3, generator=g) * 8
+        mask = torch.ones(b, 12).bool()
+        buckets = get_bucketed_distance_matrix(coords, mask)
+        assert buckets.min() >= 0 and buckets.max() <= 36
+        # symmetry
+        assert torch.equal(buckets, buckets.transpose(1, 2))
