@@ -340,7 +340,8 @@ class Alphafold2(nn.Module):
             x_point = x.reshape(b * n * n, 1, self.dim)
             t_point = t.permute(0, 2, 3, 1, 4).reshape(
                 b * n * n, num_templates, self.dim)
-            x_mask_point = x_mask.reshape(b * n * n, 1)
+            x_mask_point = x_mask.reshape(b * n * n, 1) \
+                if exists(x_mask) else None
             t_mask_point = t_mask_crossed.permute(0, 2, 3, 1).reshape(
                 b * n * n, num_templates)
 
