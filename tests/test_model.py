@@ -41,6 +41,21 @@ def test_anglegrams():
     assert ret.omega_logits.shape == (2, 32, 32, 25)
 
 
+def test_symmetrize_omega():
+    """symmetrize_omega routes the omega head through the symmetrized
+    pair rep: logits must be transpose-symmetric."""
+    model = tiny_model(predict_angles=True, symmetrize_omega=True).eval()
+    seq = torch.randint(0, 21, (1, 16))
+    msa = torch.randint(0, 21, (1, 3, 16))
+    with torch.no_grad():
+        ret = model(seq, msa)
+    assert torch.allclose(ret.omega_logits,
+                          ret.omega_logits.transpose(1, 2), atol=1e-5)
+    # theta stays on the raw (asymmetric) pair rep
+    assert not torch.allclose(ret.theta_logits,
+                              ret.theta_logits.transpose(1, 2), atol=1e-3)
+
+
 def test_templates():
     model = tiny_model(templates_dim=32, templates_angles_feats_dim=32)
     seq = torch.randint(0, 21, (2, 16))
