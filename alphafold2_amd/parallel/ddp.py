@@ -71,7 +71,7 @@ def all_reduce_mean(t: torch.Tensor):
 
 class _Bucket:
     __slots__ = ('params', 'flat', 'numel', 'ready', 'work', 'offsets',
-                 'seen')
+                 'seen', 'ever_used')
 
     def __init__(self):
         self.params: List[torch.nn.Parameter] = []
@@ -81,6 +81,12 @@ class _Bucket:
         self.work = None
         self.offsets = {}
         self.seen = set()
+        # becomes True the first time ANY param in the bucket produces a
+        # gradient; never-used buckets (e.g. template modules in a
+        # template-less run) skip their all-reduce entirely.  Monotonic,
+        # so ranks stay collective-consistent under the standard DDP
+        # contract (all ranks execute the same graph).
+        self.ever_used = False
 
 
 class DataParallelEngine:
@@ -173,6 +179,7 @@ class DataParallelEngine:
         off = bucket.offsets[p]
         bucket.flat[off:off + p.numel()].copy_(p.grad.reshape(-1))
         bucket.seen.add(p)
+        bucket.ever_used = True
         bucket.ready += 1
         if bucket.ready == len(bucket.params):
             # one large message per bucket; async so backward keeps going
@@ -185,6 +192,13 @@ class DataParallelEngine:
             return
         inv = 1.0 / self.world_size
         for bucket in self._buckets:
+            if not bucket.ever_used:
+                # no param in this bucket has EVER produced a gradient
+                # (this step included) -> its contribution is zero on
+                # every rank; skip the collective
+                if not any(p.grad is not None for p in bucket.params):
+                    continue
+                bucket.ever_used = True
             if bucket.ready != len(bucket.params):
                 # params that never got grads this step (unused path):
                 # reduce what we have for deterministic behavior

@@ -244,3 +244,51 @@ def test_ddp_full_model_step():
     g1 = torch.as_tensor(results[1])
     assert g0.shape == g1.shape
     assert torch.allclose(g0, g1, atol=1e-6)
+
+
+def _never_used_worker(rank, port, q):
+    """A bucket whose params never produce grads must skip its
+    all-reduce (flat never allocated) while used buckets still average."""
+    from alphafold2_amd.parallel import DataParallelEngine
+    _setup(rank, WORLD, port)
+    torch.manual_seed(3)
+    model = _TwoPath()
+    engine = DataParallelEngine(model, bucket_cap_mb=0.0001)  # per-param
+
+    torch.manual_seed(rank)
+    x = torch.randn(2, 8)
+    for _ in range(2):
+        model.zero_grad(set_to_none=True)
+        model(x, use_b=False).pow(2).sum().backward()
+        engine.finalize()
+    b_buckets = [engine._param_bucket[p] for p in model.b.parameters()]
+    a_buckets = [engine._param_bucket[p] for p in model.a.parameters()]
+    q.put((rank,
+           all(bk.flat is None and not bk.ever_used for bk in b_buckets),
+           all(bk.ever_used for bk in a_buckets),
+           model.a.weight.grad.numpy().copy()))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_ddp_never_used_bucket_skipped():
+    ctx = mp.get_context('spawn')
+    q = ctx.SimpleQueue()
+    port = 29519
+    procs = [ctx.Process(target=_never_used_worker, args=(r, port, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, b_skipped, a_used, a_grad = q.get()
+        results[rank] = (b_skipped, a_used, a_grad)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for r in range(WORLD):
+        assert results[r][0], 'unused bucket must never allocate/reduce'
+        assert results[r][1]
+    assert torch.allclose(torch.as_tensor(results[0][2]),
+                          torch.as_tensor(results[1][2]), atol=1e-6)
