@@ -69,3 +69,20 @@ def test_reversible_memory_constant_graph():
     assert xo.grad_fn is not None
     (xo.sum() + mo.sum()).backward()
     assert x.grad is not None and m.grad is not None
+
+
+def test_reversible_dim384_config():
+    """BASELINE configs[2] shape class (dim=384, reversible): forward +
+    backward at a reduced depth/length on CPU."""
+    from alphafold2_amd import Alphafold2
+    torch.manual_seed(0)
+    model = Alphafold2(dim=384, depth=2, heads=6, dim_head=64,
+                       reversible=True)
+    model.train()
+    seq = torch.randint(0, 21, (1, 12))
+    msa = torch.randint(0, 21, (1, 3, 12))
+    ret = model(seq, msa)
+    (ret.distance.pow(2).mean() + ret.msa_mlm_loss).backward()
+    grads = [p.grad for p in model.net.parameters() if p.grad is not None]
+    assert len(grads) > 0
+    assert all(torch.isfinite(g).all() for g in grads)
