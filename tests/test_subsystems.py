@@ -243,6 +243,32 @@ def test_native_relaxer():
 
 
 @pytest.mark.timeout(300)
+def test_predict_cli_a3m(tmp_path):
+    """predict.py --a3m: MSA file in -> PDB out (checkpoint round-trip)."""
+    import subprocess, sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    a3m = tmp_path / 'q.a3m'
+    a3m.write_text('>query\nMKTAYIAKQRQISFVK\n'
+                   '>hit1\nMKTAYIAKQRQISFVK\n'
+                   '>hit2\nMKTAYIAKQRlQISFVK\n')  # insertion removed
+    # save a checkpoint with the same tiny config to exercise --checkpoint
+    from alphafold2_amd import Alphafold2
+    from alphafold2_amd.runtime import save_checkpoint
+    ckpt = str(tmp_path / 'm.pt')
+    save_checkpoint(ckpt, Alphafold2(dim=32, depth=1, heads=8, dim_head=64,
+                                     predict_coords=True))
+    out_pdb = str(tmp_path / 'pred.pdb')
+    r = subprocess.run(
+        [sys.executable, 'predict.py', '--a3m', str(a3m),
+         '--checkpoint', ckpt,
+         '--dim', '32', '--depth', '1', '--recycles', '1',
+         '--out', out_pdb],
+        cwd=root, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert open(out_pdb).read().count('ATOM') == 16
+
+
+@pytest.mark.timeout(300)
 def test_predict_cli(tmp_path):
     """predict.py: sequence in -> CA-trace PDB with confidence out."""
     import subprocess, sys
