@@ -25,9 +25,11 @@ from .. import ops
 from ..ops.fused_modules import FusedLayerNorm
 from ..geometry.backend import torch_default_dtype
 from ..mlm import MLM
-from .evoformer import (
-    Always, Attention, Evoformer, PairwiseAttentionBlock, default, exists,
-    init_zero_,
+from .evoformer import (  # noqa: F401 -- module-level surface parity
+    Always, Attention, AxialAttention, Evoformer, EvoformerBlock,
+    FeedForward, GEGLU, MsaAttentionBlock, OuterMean,
+    PairwiseAttentionBlock, TriangleMultiplicativeModule, cast_tuple,
+    default, exists, init_zero_,
 )
 from .ipa import IPABlock
 from .quaternion import quaternion_multiply, quaternion_to_matrix

@@ -26,6 +26,10 @@ def default(val, d):
     return d() if callable(d) else d
 
 
+def cast_tuple(val, depth=1):
+    return val if isinstance(val, tuple) else (val,) * depth
+
+
 def init_zero_(layer):
     nn.init.constant_(layer.weight, 0.)
     if exists(layer.bias):

@@ -4,7 +4,7 @@ against the reference (`from alphafold2_pytorch.utils import *` style)
 ports with only the package name changed.
 """
 from .vocab import (  # noqa: F401
-    VOCAB, ProteinVocabulary, ATOM_IDS, CUSTOM_INFO,
+    VOCAB, ProteinVocabulary, ATOM_IDS, CUSTOM_INFO, get_atom_ids_dict,
     make_cloud_mask, make_atom_id_embedds, ONE_TO_THREE_LETTER_MAP,
 )
 from .geometry import (  # noqa: F401

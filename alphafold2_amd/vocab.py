@@ -112,6 +112,12 @@ def _build_atom_id_table():
 ATOM_IDS = _build_atom_id_table()
 
 
+def get_atom_ids_dict():
+    """Dict mapping each distinct atom name to a token id (reference
+    utils.py:108-116 parity; '' is the padding slot)."""
+    return dict(ATOM_IDS)
+
+
 def make_cloud_mask(aa: str):
     """(14,) float mask: 1 for occupied atom slots of this residue type."""
     import numpy as np
