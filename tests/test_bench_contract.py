@@ -14,6 +14,13 @@ TINY = ['--dim', '32', '--depth', '1', '--crop-len', '32',
         '--steps', '2', '--warmup', '1']
 
 
+def _free_port():
+    import socket
+    with socket.socket() as s:
+        s.bind(('127.0.0.1', 0))
+        return s.getsockname()[1]
+
+
 def _parse_last_json(out):
     for line in reversed(out.strip().splitlines()):
         if line.startswith('{'):
@@ -43,7 +50,7 @@ def test_bench_torchrun_two_ranks_cpu():
     out = subprocess.run(
         [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
          '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
-         '--master-port', '29531', 'bench.py', '--gpus', '2'] + TINY,
+         '--master-port', str(_free_port()), 'bench.py', '--gpus', '2'] + TINY,
         cwd=ROOT, capture_output=True, text=True, timeout=280, env=env)
     assert out.returncode == 0, out.stderr[-2000:]
     d = _parse_last_json(out.stdout)

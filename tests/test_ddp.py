@@ -13,6 +13,15 @@ from alphafold2_amd.models.evoformer import FeedForward
 WORLD = 2
 
 
+def _free_port():
+    """OS-assigned free port (fixed ports collide with TIME_WAIT when
+    suites run back-to-back)."""
+    import socket
+    with socket.socket() as s:
+        s.bind(('127.0.0.1', 0))
+        return s.getsockname()[1]
+
+
 def _setup(rank, world_size, port):
     os.environ['MASTER_ADDR'] = '127.0.0.1'
     os.environ['MASTER_PORT'] = str(port)
@@ -47,7 +56,7 @@ def _ddp_worker(rank, port, q):
 def test_ddp_grad_allreduce():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
-    port = 29511
+    port = _free_port()
     procs = [ctx.Process(target=_ddp_worker, args=(r, port, q))
              for r in range(WORLD)]
     for p in procs:
@@ -107,7 +116,7 @@ def _no_sync_worker(rank, port, q):
 def test_ddp_no_sync_accumulation():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
-    port = 29513
+    port = _free_port()
     procs = [ctx.Process(target=_no_sync_worker, args=(r, port, q))
              for r in range(WORLD)]
     for p in procs:
@@ -161,7 +170,7 @@ def _unused_param_worker(rank, port, q):
 def test_ddp_unused_param_in_sync_micro():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
-    port = 29515
+    port = _free_port()
     procs = [ctx.Process(target=_unused_param_worker, args=(r, port, q))
              for r in range(WORLD)]
     for p in procs:
@@ -227,7 +236,7 @@ def _model_worker(rank, port, q):
 def test_ddp_full_model_step():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
-    port = 29517
+    port = _free_port()
     procs = [ctx.Process(target=_model_worker, args=(r, port, q))
              for r in range(WORLD)]
     for p in procs:
@@ -275,7 +284,7 @@ def _never_used_worker(rank, port, q):
 def test_ddp_never_used_bucket_skipped():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
-    port = 29519
+    port = _free_port()
     procs = [ctx.Process(target=_never_used_worker, args=(r, port, q))
              for r in range(WORLD)]
     for p in procs:
