@@ -255,7 +255,7 @@ except ImportError:  # pragma: no cover
     HAS_HYP = False
 
 if HAS_HYP:
-    @settings(max_examples=25, deadline=None)
+    @settings(max_examples=25, deadline=None, derandomize=True)
     @given(st.integers(min_value=4, max_value=64), st.integers(0, 10**6))
     def test_kabsch_rigid_motion_invariant(n, seed):
         """Kabsch must recover ANY rigid motion: RMSD(after) ~ 0."""
@@ -270,7 +270,7 @@ if HAS_HYP:
         a_, b_ = Kabsch(a, b)
         assert RMSD(a_, b_).item() < 1e-6
 
-    @settings(max_examples=25, deadline=None)
+    @settings(max_examples=25, deadline=None, derandomize=True)
     @given(st.integers(min_value=5, max_value=40), st.integers(0, 10**6))
     def test_tmscore_bounds_and_self(n, seed):
         g = torch.Generator().manual_seed(seed)
@@ -280,7 +280,7 @@ if HAS_HYP:
         assert 0.0 <= tm.item() <= 1.0
         assert torch.allclose(TMscore(a, a), torch.ones(1))
 
-    @settings(max_examples=15, deadline=None)
+    @settings(max_examples=15, deadline=None, derandomize=True)
     @given(st.integers(min_value=6, max_value=24), st.integers(0, 10**6))
     def test_lddt_bounds(n, seed):
         g = torch.Generator().manual_seed(seed)
@@ -290,7 +290,7 @@ if HAS_HYP:
         val = lddt_ca_torch(a, b, cloud)
         assert (val >= 0).all() and (val <= 1).all()
 
-    @settings(max_examples=15, deadline=None)
+    @settings(max_examples=15, deadline=None, derandomize=True)
     @given(st.integers(min_value=2, max_value=8), st.integers(0, 10**6))
     def test_distogram_bucket_bounds(b, seed):
         from alphafold2_amd.utils import get_bucketed_distance_matrix
