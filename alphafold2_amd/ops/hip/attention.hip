@@ -6,8 +6,10 @@
 // (batch b*n, len n, pair bias), and template pointwise attention.
 //
 // MI355X-first design decisions:
-//  * MFMA bf16 16x16x32 tiles; fp32 accumulation; one workgroup = 4 waves
-//    = 64 query rows; KV tiled by 64 with LDS staging.
+//  * MFMA bf16 16x16x32 tiles; fp32 accumulation.  Forward: one
+//    workgroup = 8 waves = 128 query rows (FBQ/FNT); backward kernels
+//    run 4-wave workgroups over 64-row tiles (BQ/NWAVES).  KV tiled by
+//    64 with double-buffered LDS staging (fwd: single barrier/tile).
 //  * The pair bias is NOT materialized per folded axis: the kernel takes
 //    bias of shape (B / bias_repeat, h, Lq, Lk) and folds the repeat in
 //    the index — the eager path would replicate it axial_dim times
