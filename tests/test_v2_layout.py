@@ -173,3 +173,212 @@ def test_v2_dataflow_matches_attention():
             out[row, c * 16 + myq] = o[c, :, reg] / l_row
     assert np.allclose(out, O_ref, atol=1e-9), \
         np.abs(out - O_ref).max()
+
+
+def _exchange_to_afrag(vals, g, myq):
+    """The cvt_pk+shfl C->A exchange shared by all v2 kernels.
+
+    vals: (4, NL, 4) per-lane C-layout tiles where vals[c, lane, reg]
+    = M[x = c*16 + g*4 + reg][y = myq] for some matrix M.
+    Returns (2, NL, 8) A-operand fragments a[kblk, lane, j]
+    = M[x = g*8 + j + 32*kblk][y = myq].
+    """
+    out = np.zeros((2, NL, 8))
+    for kblk in range(2):
+        ca, cb = 2 * kblk, 2 * kblk + 1
+        packs = {
+            ('a', 0): vals[ca, :, 0:2], ('a', 1): vals[ca, :, 2:4],
+            ('b', 0): vals[cb, :, 0:2], ('b', 1): vals[cb, :, 2:4],
+        }
+        src_lo = myq + 16 * ((2 * g) & 3)
+        src_hi = myq + 16 * ((2 * g + 1) & 3)
+        for which, src in (('lo', src_lo), ('hi', src_hi)):
+            base = 0 if which == 'lo' else 4
+            for half in range(2):
+                got_a = shfl(packs[('a', half)], src)
+                got_b = shfl(packs[('b', half)], src)
+                use_b = (g >= 2)[:, None]
+                out[kblk, :, base + 2 * half: base + 2 * half + 2] = \
+                    np.where(use_b, got_b, got_a)
+    return out
+
+
+def test_v2_dq_dataflow():
+    """v2 dQ: swapped-S recompute (per-lane lse/delta scalars, NO
+    reductions), dS^T in registers, exchange -> mfma(dS, K^T)."""
+    rng = np.random.default_rng(1)
+    Q = rng.standard_normal((NQ, DH))
+    K = rng.standard_normal((NKV, DH))
+    V = rng.standard_normal((NKV, DH))
+    dO = rng.standard_normal((NQ, DH))
+    bias = rng.standard_normal((NQ, NKV))
+    scale = DH ** -0.5
+
+    # reference backward pieces
+    S_ref = Q @ K.T * scale + bias
+    m = S_ref.max(axis=1, keepdims=True)
+    P_unn = np.exp(S_ref - m)
+    l = P_unn.sum(axis=1, keepdims=True)
+    P = P_unn / l
+    O = P @ V
+    lse = (m + np.log(l))[:, 0]
+    delta = (dO * O).sum(axis=1)
+    dP = dO @ V.T
+    dS = P * (dP - delta[:, None])
+    dQ_ref = scale * dS @ K
+
+    lanes = np.arange(NL)
+    g = lanes >> 4
+    myq = lanes & 15
+
+    qf = [q_fragment(Q, dblk) for dblk in range(2)]
+    dof = [q_fragment(dO, dblk) for dblk in range(2)]
+
+    # swapped S^T and dP^T recompute
+    s = np.stack([sum(mfma_16x16x32(k_fragment(K, c, dblk), qf[dblk])
+                      for dblk in range(2)) for c in range(4)])
+    dp = np.stack([sum(mfma_16x16x32(k_fragment(V, c, dblk), dof[dblk])
+                       for dblk in range(2)) for c in range(4)])
+
+    # per-lane: p = exp(s*scale + bias - lse[myq]); ds = p*(dp - delta[myq])
+    ds = np.zeros_like(s)
+    for c in range(4):
+        for reg in range(4):
+            kv = c * 16 + g * 4 + reg
+            p_l = np.exp(s[c, :, reg] * scale + bias[myq, kv] - lse[myq])
+            ds[c, :, reg] = p_l * (dp[c, :, reg] - delta[myq])
+            assert np.allclose(ds[c, :, reg], dS[myq, kv], atol=1e-9)
+
+    # exchange -> A fragments of dS (row=q, k=kv); B = K^T fragments
+    ds_frag = _exchange_to_afrag(ds, g, myq)
+    dq_acc = np.zeros((4, NL, 4))
+    for c in range(4):
+        for kblk in range(2):
+            dq_acc[c] += mfma_16x16x32(ds_frag[kblk],
+                                       vt_fragment(K, c, kblk))
+
+    dq_out = np.zeros((NQ, DH))
+    for reg in range(4):
+        for c in range(4):
+            dq_out[g * 4 + reg, c * 16 + myq] = dq_acc[c, :, reg] * scale
+    assert np.allclose(dq_out, dQ_ref, atol=1e-9), \
+        np.abs(dq_out - dQ_ref).max()
+
+
+def test_v2_dkv_dataflow():
+    """v2 dK/dV: STANDARD-orientation S (mfma(Q,K) -> D[q][kv], lane
+    holds one kv column), per-lane P/dS from lse/delta of the lane's 4
+    q rows, then the SAME exchange turns the P^T / dS^T column layout
+    into A-fragments (row=kv, k=q) -> mfma against dO / Q B-fragments.
+    Replaces the production dkv kernel's two LDS round-trips."""
+    rng = np.random.default_rng(2)
+    NQT = 32  # q rows processed per iteration (k=32 for the mfma)
+    Q = rng.standard_normal((NQT, DH))
+    K = rng.standard_normal((NKV, DH))  # block's kv tile: 64 rows
+    V = rng.standard_normal((NKV, DH))
+    dO = rng.standard_normal((NQT, DH))
+    bias = rng.standard_normal((NQT, NKV))
+    scale = DH ** -0.5
+
+    S_ref = Q @ K.T * scale + bias
+    # lse/delta computed over the FULL kv length; single tile here
+    m = S_ref.max(axis=1, keepdims=True)
+    P_unn = np.exp(S_ref - m)
+    l = P_unn.sum(axis=1, keepdims=True)
+    P = P_unn / l
+    lse = (m + np.log(l))[:, 0]
+    delta = (dO * (P @ V)).sum(axis=1)
+    dP = dO @ V.T
+    dS = P * (dP - delta[:, None])
+    dV_ref = P.T @ dO
+    dK_ref = scale * dS.T @ Q
+
+    lanes = np.arange(NL)
+    g = lanes >> 4
+    mykv = lanes & 15  # for the standard orientation, col = kv
+
+    # standard S: two q-16 tiles -> (2 qtiles, 4 kvtiles) of C layout;
+    # arrange as vals[chunk][lane][reg] with chunk = qtile*... the
+    # exchange wants M[x][y] with x = c*16+g*4+reg = q, y = mykv: that
+    # is exactly C-layout D[q][kv] of mfma(Q-tile c, K), c = q-tile
+    # index 0..1 for 32 q -> pad chunks 2,3 with a second 16-kv... here
+    # x runs over q (32) so chunks c=0,1 hold q tiles; the A-fragment
+    # k-axis is q (32 = one kblk).  Use a 2-chunk variant directly.
+    p_cols = np.zeros((2, NL, 4))   # P^T column layout per q-tile
+    ds_cols = np.zeros((2, NL, 4))
+    for qt in range(2):
+        qf = [q_fragment(Q[qt * 16:(qt + 1) * 16], dblk) for dblk in range(2)]
+        acc = np.zeros((NL, 4))
+        for dblk in range(2):
+            # standard orientation: A = Q rows, B = K rows as col axis
+            acc += mfma_16x16x32(qf[dblk], kb_fragment(K, dblk))
+        for reg in range(4):
+            q_idx = qt * 16 + g * 4 + reg
+            s_val = acc[:, reg] * scale + bias[q_idx, mykv]
+            assert np.allclose(acc[:, reg], (Q @ K.T)[q_idx, mykv],
+                               atol=1e-9)
+            p_l = np.exp(s_val - lse[q_idx])
+            p_cols[qt, :, reg] = p_l
+            dp_l = (dO @ V.T)[q_idx, mykv]
+            ds_cols[qt, :, reg] = p_l * (dp_l - delta[q_idx])
+
+    # exchange: target a[j] = M^T[kv = lane&15][q = g*8 + j], q-chunks
+    # {2g, 2g+1} live in (qtile = chunk>>2, group = chunk&3)
+    def exchange_cols(cols):
+        out = np.zeros((NL, 8))
+        packs = {(qt, half): cols[qt, :, 2 * half:2 * half + 2]
+                 for qt in range(2) for half in range(2)}
+        src_lo = mykv + 16 * ((2 * g) & 3)
+        src_hi = mykv + 16 * ((2 * g + 1) & 3)
+        for which, src in (('lo', src_lo), ('hi', src_hi)):
+            base = 0 if which == 'lo' else 4
+            for half in range(2):
+                got0 = shfl(packs[(0, half)], src)
+                got1 = shfl(packs[(1, half)], src)
+                use1 = (g >= 2)[:, None]
+                out[:, base + 2 * half: base + 2 * half + 2] = \
+                    np.where(use1, got1, got0)
+        return out
+
+    p_frag = exchange_cols(p_cols)
+    ds_frag = exchange_cols(ds_cols)
+    for j in range(8):
+        assert np.allclose(p_frag[:, j], P.T[mykv, g * 8 + j], atol=1e-9)
+
+    dv_acc = np.zeros((4, NL, 4))
+    dk_acc = np.zeros((4, NL, 4))
+
+    # B-fragments over the q k-axis: b[j] = X[q=(lane>>4)*8+j][col]
+    def bfrag_q(X, ctile):
+        regs = np.zeros((NL, 8))
+        for lane in range(NL):
+            for j in range(8):
+                regs[lane, j] = X[(lane >> 4) * 8 + j,
+                                  ctile * 16 + (lane & 15)]
+        return regs
+
+    for c in range(4):
+        dv_acc[c] += mfma_16x16x32(p_frag, bfrag_q(dO, c))
+        dk_acc[c] += mfma_16x16x32(ds_frag, bfrag_q(Q, c))
+
+    dv_out = np.zeros((NKV, DH))
+    dk_out = np.zeros((NKV, DH))
+    for reg in range(4):
+        for c in range(4):
+            dv_out[g * 4 + reg, c * 16 + mykv] = dv_acc[c, :, reg]
+            dk_out[g * 4 + reg, c * 16 + mykv] = dk_acc[c, :, reg] * scale
+    # the block covers kv rows 0..15 per wave-tile here (single 16-kv
+    # tile simulated); restrict the comparison accordingly
+    assert np.allclose(dv_out[:16], dV_ref[:16], atol=1e-9), \
+        np.abs(dv_out[:16] - dV_ref[:16]).max()
+    assert np.allclose(dk_out[:16], dK_ref[:16], atol=1e-9)
+
+
+def kb_fragment(K, dblk):
+    """B-operand for standard-orientation QK^T: col = K row (kv),
+    k = dh slice — b[j] = K[col=lane&15][k=dblk*32+(lane>>4)*8+j]."""
+    regs = np.zeros((NL, 8))
+    for lane in range(NL):
+        for j in range(8):
+            regs[lane, j] = K[lane & 15, dblk * 32 + (lane >> 4) * 8 + j]
+    return regs
