@@ -6,8 +6,20 @@
 //     && /tmp/v2
 //
 // prints PASS/FAIL vs a CPU fp32 reference for {bias,mask} x shapes,
-// plus a TF rate at the production triangle-attention shape to compare
-// against tools/attn_bench.py (fwd ~190 TF at end of round 1).
+// plus TF rates at the production triangle-attention shape to compare
+// against tools/attn_bench.py (end of round 1: fwd ~190 TF, dq
+// 1.37 ms, dkv 4.0 ms per call at batch 5).
+//
+// Contains all THREE kernels of the v2 design (every dataflow
+// machine-verified on CPU by tests/test_v2_layout.py):
+//   attn_fwd_v2   94 VGPRs / 48KB LDS   (prod: 128 / 64KB)
+//   attn_dq_v2   126 VGPRs / 80KB LDS   (prod: 178, 12 waves/CU -> 16)
+//   attn_dkv_v2  122 VGPRs / 64KB LDS   (prod: 249, 8 waves/CU -> 16)
+// Known prototype gaps vs production semantics: no bias_repeat fold
+// (pass pre-expanded bias), no dbias accumulation in dkv (the
+// production drain logic ports unchanged — dS is available per-lane at
+// the marked spot), and dkv's per-lane bias loads should be LDS-staged
+// in the production port.
 //
 // Key derivation (verified on paper against the HW-probed fragment
 // layouts in tools/mfma_probe.hip — C/D: col=lane&15,
@@ -667,6 +679,224 @@ static int run_case(int B, int Lq, int Lk, bool has_bias, bool has_mask,
   return pass ? 0 : 1;
 }
 
+// ---- v2 backward-dK/dV kernel (dataflow CPU-verified by
+// tests/test_v2_layout.py::test_v2_dkv_dataflow) ---------------------------
+// STANDARD orientation: block owns BKV=128 kv rows (8 waves x 16), q
+// streams through in 32-row chunks.  Per lane: kv = wave*16 + (lane&15)
+// fixed; S/dP arrive in C layout [q][kv-col]; P and dS are formed
+// per-lane from lse/delta, then the SAME cvt_pk+shfl exchange used by
+// the forward turns their column layout into A-fragments (row=kv,
+// k=q32) for the dV/dK MFMAs — replacing the production kernel's two
+// LDS round-trips.  dBias is omitted in the prototype (the production
+// drain logic ports unchanged: dS is available per-lane right here).
+// B-fragment over a 32-long q k-axis from a [dh][q] transposed tile
+__device__ __forceinline__ bf16x8 fragq_row32(const char* lds, int dhrow) {
+  const int lane = threadIdx.x & 63;
+  return *reinterpret_cast<const bf16x8*>(
+      lds + dhrow * 64 + ((lane >> 4) << 4));
+}
+
+constexpr int BKV = 128;   // kv rows per block
+constexpr int QC = 32;     // q rows per iteration (mfma k-axis)
+
+template <bool HAS_BIAS, bool HAS_MASK>
+__global__ __launch_bounds__(FNT, 4)
+void attn_dkv_v2(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+                 const bf16_t* __restrict__ v,
+                 const bf16_t* __restrict__ bias,
+                 const unsigned char* __restrict__ mask,
+                 const bf16_t* __restrict__ dout,
+                 const float* __restrict__ lse,
+                 const float* __restrict__ delta,
+                 bf16_t* __restrict__ dk_out, bf16_t* __restrict__ dv_out,
+                 int B, int Lq, int Lk, float scale) {
+  __shared__ char k_lds[BKV * ROWB];      // block's kv tile, row-major
+  __shared__ char v_lds[BKV * ROWB];
+  __shared__ char q_lds[2][QC * ROWB];    // q chunk, row-major (S A-op)
+  __shared__ char do_lds[2][QC * ROWB];   // dO chunk, row-major (dP A-op)
+  __shared__ char qt_lds[2][DH * (QC * 2)];   // Q^T  [dh][q] (dK B-op)
+  __shared__ char dot_lds[2][DH * (QC * 2)];  // dO^T [dh][q] (dV B-op)
+
+  const int kvtile = blockIdx.x;
+  const int batch = blockIdx.y;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int g = lane >> 4;
+  const int mykv = lane & 15;                // kv col within the wave tile
+  const int kv_row = wave * 16 + mykv;       // within the block tile
+
+  const bf16_t* k_g = k + ((long)batch * Lk + (long)kvtile * BKV) * DH;
+  const bf16_t* v_g = v + ((long)batch * Lk + (long)kvtile * BKV) * DH;
+  const bf16_t* q_g = q + (long)batch * Lq * DH;
+  const bf16_t* do_g = dout + (long)batch * Lq * DH;
+  const float* lse_g = lse + (long)batch * Lq;
+  const float* dl_g = delta + (long)batch * Lq;
+  const bf16_t* bias_g = HAS_BIAS
+      ? bias + (long)batch * Lq * Lk + (long)kvtile * BKV : nullptr;
+
+  const int kv_rows = min(BKV, Lk - kvtile * BKV);
+  const bool kv_ok = kv_row < kv_rows &&
+      (!HAS_MASK ||
+       mask[(long)batch * Lk + kvtile * BKV + kv_row]);
+
+  {  // stage the block's K/V tiles once (128 rows each)
+    StageRegs<BKV> kr, vr;
+    stage_load<BKV>(k_g, DH, kv_rows, kr);
+    stage_load<BKV>(v_g, DH, kv_rows, vr);
+    stage_store<BKV>(kr, k_lds);
+    stage_store<BKV>(vr, v_lds);
+  }
+  __syncthreads();
+
+  f32x4 dk_acc[4], dv_acc[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    dk_acc[c] = f32x4{0, 0, 0, 0};
+    dv_acc[c] = f32x4{0, 0, 0, 0};
+  }
+
+  // stage a 32-row q/dO chunk in BOTH layouts: threads 0..255 move the
+  // q tile, 256..511 the dO tile (32 rows x 8 16B chunks each).  The
+  // row-major copy is swizzled (frag_row applies swz on reads); the
+  // transposed [dh][q] copy has 64B rows — a b128 read of 16 rows hits
+  // the 8-slot LDS floor with or without a swizzle, so it stays plain.
+  auto stage_chunk = [&](const bf16_t* gq, const bf16_t* gdo, int rows,
+                         int buf) {
+    const int half = threadIdx.x >> 8;
+    const int idx = threadIdx.x & 255;
+    const int row = idx >> 3;            // 0..31
+    const int c16 = (idx & 7) << 4;      // 16B chunk in the 128B row
+    const bf16_t* gsrc = half ? gdo : gq;
+    char* rm = half ? do_lds[buf] : q_lds[buf];
+    char* tr = half ? dot_lds[buf] : qt_lds[buf];
+    float4 val = {0, 0, 0, 0};
+    if (row < rows)
+      val = *reinterpret_cast<const float4*>(
+          reinterpret_cast<const char*>(gsrc + row * DH) + c16);
+    *reinterpret_cast<float4*>(rm + swz(row, c16)) = val;
+    const bf16_t* vv = reinterpret_cast<const bf16_t*>(&val);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int dh = (c16 >> 1) + j;
+      *reinterpret_cast<bf16_t*>(
+          tr + dh * (QC * 2) + row * (int)sizeof(bf16_t)) = vv[j];
+    }
+  };
+
+  const int n_qc = (Lq + QC - 1) / QC;
+  stage_chunk(q_g, do_g, min(QC, Lq), 0);
+  __syncthreads();
+
+  for (int qc = 0; qc < n_qc; ++qc) {
+    const int buf = qc & 1;
+    const int q_base = qc * QC;
+    const int q_rows = min(QC, Lq - q_base);
+    if (qc + 1 < n_qc) {
+      stage_chunk(q_g + (long)(qc + 1) * QC * DH,
+                  do_g + (long)(qc + 1) * QC * DH,
+                  min(QC, Lq - (qc + 1) * QC), 1 - buf);
+    }
+    // NOTE prototype keeps 1 barrier before compute; the prefetch above
+    // writes the OTHER buffer so this is the same single-barrier scheme
+    __syncthreads();
+
+    // S and dP chunks in C layout: per lane [q=16qt+g*4+reg][kv=mykv]
+    f32x4 s_cols[2], dp_cols[2];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int qt = 0; qt < 2; ++qt) {
+      f32x4 sa = {0, 0, 0, 0}, da = {0, 0, 0, 0};
+#pragma unroll
+      for (int dblk = 0; dblk < 2; ++dblk) {
+        // A = Q / dO rows (16 q), B = K / V rows as columns (16 kv)
+        bf16x8 qa = frag_row(q_lds[buf], qt * 16 + mykv, dblk);
+        bf16x8 doa = frag_row(do_lds[buf], qt * 16 + mykv, dblk);
+        bf16x8 kb = frag_row(k_lds, wave * 16 + mykv, dblk);
+        bf16x8 vb = frag_row(v_lds, wave * 16 + mykv, dblk);
+        sa = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa, kb, sa, 0, 0, 0);
+        da = __builtin_amdgcn_mfma_f32_16x16x32_bf16(doa, vb, da, 0, 0, 0);
+      }
+      s_cols[qt] = sa;
+      dp_cols[qt] = da;
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // per-lane P and dS columns
+    f32x4 p_cols[2], ds_cols[2];
+#pragma unroll
+    for (int qt = 0; qt < 2; ++qt) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int qi = q_base + qt * 16 + g * 4 + reg;
+        const bool q_in = qi < Lq && (qt * 16 + g * 4 + reg) < q_rows;
+        float val = s_cols[qt][reg] * scale;
+        if (HAS_BIAS && q_in && kv_ok)
+          val += (float)bias_g[(long)qi * Lk + kv_row];
+        const float lse_l = q_in ? lse_g[qi] : NEG_INF;
+        const float p = (kv_ok && q_in && lse_l > NEG_INF)
+            ? __expf(val - lse_l) : 0.f;
+        p_cols[qt][reg] = p;
+        ds_cols[qt][reg] = p * (dp_cols[qt][reg] - (q_in ? dl_g[qi] : 0.f));
+      }
+    }
+
+    // exchange columns -> A-fragments (row = kv, k = the 32 q rows)
+    auto exchange = [&](const f32x4* cols) -> bf16x8 {
+      const unsigned p00 = pack_bf16(cols[0][0], cols[0][1]);
+      const unsigned p01 = pack_bf16(cols[0][2], cols[0][3]);
+      const unsigned p10 = pack_bf16(cols[1][0], cols[1][1]);
+      const unsigned p11 = pack_bf16(cols[1][2], cols[1][3]);
+      const int src_lo = mykv + 16 * ((2 * g) & 3);
+      const int src_hi = mykv + 16 * ((2 * g + 1) & 3);
+      unsigned lo00 = __shfl(p00, src_lo, 64), lo01 = __shfl(p01, src_lo, 64);
+      unsigned hi00 = __shfl(p00, src_hi, 64), hi01 = __shfl(p01, src_hi, 64);
+      unsigned lo10 = __shfl(p10, src_lo, 64), lo11 = __shfl(p11, src_lo, 64);
+      unsigned hi10 = __shfl(p10, src_hi, 64), hi11 = __shfl(p11, src_hi, 64);
+      union {
+        unsigned u[4];
+        bf16x8 f;
+      } af;
+      const bool use1 = g >= 2;   // q chunks {2g,2g+1}: qtile = chunk>=4
+      af.u[0] = use1 ? lo10 : lo00;
+      af.u[1] = use1 ? lo11 : lo01;
+      af.u[2] = use1 ? hi10 : hi00;
+      af.u[3] = use1 ? hi11 : hi01;
+      return af.f;
+    };
+    bf16x8 p_frag = exchange(p_cols);
+    bf16x8 ds_frag = exchange(ds_cols);
+
+    // dV += P^T dO ; dK += dS^T Q   (B = transposed chunks, k = q)
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 dob = fragq_row32(dot_lds[buf], c * 16 + mykv);
+      bf16x8 qb = fragq_row32(qt_lds[buf], c * 16 + mykv);
+      dv_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          p_frag, dob, dv_acc[c], 0, 0, 0);
+      dk_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          ds_frag, qb, dk_acc[c], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  // epilogue: rows kv' = wave*16 + g*4 + reg
+  bf16_t* dk_g = dk_out + ((long)batch * Lk + (long)kvtile * BKV) * DH;
+  bf16_t* dv_g = dv_out + ((long)batch * Lk + (long)kvtile * BKV) * DH;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = wave * 16 + g * 4 + reg;
+    if (row < kv_rows) {
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        dk_g[(long)row * DH + c * 16 + mykv] =
+            (bf16_t)(dk_acc[c][reg] * scale);
+        dv_g[(long)row * DH + c * 16 + mykv] = (bf16_t)dv_acc[c][reg];
+      }
+    }
+  }
+}
+
 static int run_dq_case(int B, int Lq, int Lk, bool has_bias,
                        bool timing) {
   const float scale = 1.f / sqrtf((float)DH);
@@ -785,6 +1015,144 @@ static int run_dq_case(int B, int Lq, int Lk, bool has_bias,
   return pass ? 0 : 1;
 }
 
+static int run_dkv_case(int B, int Lq, int Lk, bool has_bias,
+                        bool has_mask, bool timing) {
+  const float scale = 1.f / sqrtf((float)DH);
+  srand(4242);
+  auto rnd = [&]() { return (rand() / (float)RAND_MAX - 0.5f) * 2.f; };
+
+  std::vector<float> qf((long)B * Lq * DH), kf((long)B * Lk * DH),
+      vf((long)B * Lk * DH), dof((long)B * Lq * DH),
+      bf(has_bias ? (long)B * Lq * Lk : 1);
+  std::vector<unsigned char> mk(has_mask ? (long)B * Lk : 1, 1);
+  for (auto& x : qf) x = rnd();
+  for (auto& x : kf) x = rnd();
+  for (auto& x : vf) x = rnd();
+  for (auto& x : dof) x = rnd();
+  for (auto& x : bf) x = rnd();
+  if (has_mask)
+    for (size_t i = 0; i < mk.size(); ++i)
+      mk[i] = (i % Lk == 0) ? 1 : (rnd() > -0.6f);
+
+  auto to_bf = [](std::vector<float>& s) {
+    std::vector<bf16_t> o(s.size());
+    for (size_t i = 0; i < s.size(); ++i) {
+      o[i] = (bf16_t)s[i];
+      s[i] = (float)o[i];
+    }
+    return o;
+  };
+  auto qb = to_bf(qf), kb = to_bf(kf), vb = to_bf(vf), dob = to_bf(dof),
+       bb = to_bf(bf);
+
+  // CPU: lse/delta + reference dK/dV
+  std::vector<float> lse((long)B * Lq), delta((long)B * Lq),
+      dk_ref((long)B * Lk * DH, 0.f), dv_ref((long)B * Lk * DH, 0.f);
+  std::vector<float> srow(Lk), prow(Lk);
+  for (int b = 0; b < B; ++b)
+    for (int i = 0; i < Lq; ++i) {
+      float mx = -1e30f;
+      for (int j = 0; j < Lk; ++j) {
+        bool ok = !has_mask || mk[(long)b * Lk + j];
+        float s = 0.f;
+        for (int d = 0; d < DH; ++d)
+          s += qf[((long)b * Lq + i) * DH + d] *
+               kf[((long)b * Lk + j) * DH + d];
+        s *= scale;
+        if (has_bias && ok) s += bf[((long)b * Lq + i) * Lk + j];
+        srow[j] = ok ? s : -1e30f;
+        mx = fmaxf(mx, srow[j]);
+      }
+      float l = 0.f;
+      for (int j = 0; j < Lk; ++j) {
+        prow[j] = (mx <= -1e30f) ? 0.f : expf(srow[j] - mx);
+        l += prow[j];
+      }
+      lse[(long)b * Lq + i] = (l > 0.f) ? mx + logf(l) : -1e30f;
+      float dl = 0.f;
+      for (int d = 0; d < DH; ++d) {
+        float o = 0.f;
+        for (int j = 0; j < Lk; ++j)
+          o += (l > 0 ? prow[j] / l : 0.f) *
+               vf[((long)b * Lk + j) * DH + d];
+        dl += o * dof[((long)b * Lq + i) * DH + d];
+      }
+      delta[(long)b * Lq + i] = dl;
+      for (int j = 0; j < Lk; ++j) {
+        const float p = l > 0 ? prow[j] / l : 0.f;
+        float dp = 0.f;
+        for (int d = 0; d < DH; ++d)
+          dp += dof[((long)b * Lq + i) * DH + d] *
+                vf[((long)b * Lk + j) * DH + d];
+        const float ds = p * (dp - dl);
+        for (int d = 0; d < DH; ++d) {
+          dv_ref[((long)b * Lk + j) * DH + d] +=
+              p * dof[((long)b * Lq + i) * DH + d];
+          dk_ref[((long)b * Lk + j) * DH + d] +=
+              scale * ds * qf[((long)b * Lq + i) * DH + d];
+        }
+      }
+    }
+
+  bf16_t *dq_ = to_dev(qb), *dk_ = to_dev(kb), *dv_ = to_dev(vb),
+         *ddo = to_dev(dob), *db_ = to_dev(bb);
+  unsigned char* dm = to_dev(mk);
+  float *dlse = to_dev(lse), *ddelta = to_dev(delta);
+  bf16_t *dko, *dvo;
+  HIP_CHECK(hipMalloc(&dko, (long)B * Lk * DH * sizeof(bf16_t)));
+  HIP_CHECK(hipMalloc(&dvo, (long)B * Lk * DH * sizeof(bf16_t)));
+
+  dim3 grid((Lk + BKV - 1) / BKV, B), block(FNT);
+#define LAUNCH_DKV(HB, HM)                                                hipLaunchKernelGGL((attn_dkv_v2<HB, HM>), grid, block, 0, 0, dq_,                          dk_, dv_, db_, dm, ddo, dlse, ddelta, dko, dvo,                         B, Lq, Lk, scale)
+  if (has_bias && has_mask) LAUNCH_DKV(true, true);
+  else if (has_bias) LAUNCH_DKV(true, false);
+  else if (has_mask) LAUNCH_DKV(false, true);
+  else LAUNCH_DKV(false, false);
+  HIP_CHECK(hipDeviceSynchronize());
+
+  std::vector<bf16_t> dkb((long)B * Lk * DH), dvb((long)B * Lk * DH);
+  HIP_CHECK(hipMemcpy(dkb.data(), dko, dkb.size() * sizeof(bf16_t),
+                      hipMemcpyDeviceToHost));
+  HIP_CHECK(hipMemcpy(dvb.data(), dvo, dvb.size() * sizeof(bf16_t),
+                      hipMemcpyDeviceToHost));
+  float err_k = 0.f, err_v = 0.f, mx_k = 0.f, mx_v = 0.f;
+  for (size_t i = 0; i < dk_ref.size(); ++i) {
+    err_k = fmaxf(err_k, fabsf((float)dkb[i] - dk_ref[i]));
+    err_v = fmaxf(err_v, fabsf((float)dvb[i] - dv_ref[i]));
+    mx_k = fmaxf(mx_k, fabsf(dk_ref[i]));
+    mx_v = fmaxf(mx_v, fabsf(dv_ref[i]));
+  }
+  const bool pass = err_k < 6e-2f * fmaxf(1.f, mx_k) &&
+                    err_v < 6e-2f * fmaxf(1.f, mx_v);
+  printf("dkv: B=%d Lq=%d Lk=%d bias=%d mask=%d  dK_err=%.4f dV_err=%.4f"
+         "  %s\n", B, Lq, Lk, has_bias, has_mask, err_k, err_v,
+         pass ? "PASS" : "FAIL");
+
+  if (timing && pass) {
+    hipEvent_t e0, e1;
+    HIP_CHECK(hipEventCreate(&e0));
+    HIP_CHECK(hipEventCreate(&e1));
+    for (int i = 0; i < 5; ++i) LAUNCH_DKV(true, false);
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipEventRecord(e0));
+    const int iters = 50;
+    for (int i = 0; i < iters; ++i) LAUNCH_DKV(true, false);
+    HIP_CHECK(hipEventRecord(e1));
+    HIP_CHECK(hipEventSynchronize(e1));
+    float ms;
+    HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+    ms /= iters;
+    const double fl = 8.0 * B * (double)Lq * Lk * DH;  // S+dP+dV+dK
+    printf("  dkv timing: %.3f ms  %.1f TF  (production dkv ~2.6 ms "
+           "at this shape, incl. dbias)\n", ms,
+           fl / (ms * 1e-3) / 1e12);
+  }
+#undef LAUNCH_DKV
+  hipFree(dq_); hipFree(dk_); hipFree(dv_); hipFree(ddo); hipFree(db_);
+  hipFree(dm); hipFree(dlse); hipFree(ddelta); hipFree(dko); hipFree(dvo);
+  return pass ? 0 : 1;
+}
+
 int main() {
   int rc = 0;
   rc |= run_case(4, 128, 128, false, false, false);
@@ -799,6 +1167,11 @@ int main() {
   rc |= run_dq_case(4, 128, 128, false, false);
   rc |= run_dq_case(3, 100, 72, true, false);
   rc |= run_dq_case(10240, 256, 256, true, true);
+  // backward-dK/dV prototype
+  rc |= run_dkv_case(4, 128, 128, false, false, false);
+  rc |= run_dkv_case(3, 100, 72, true, true, false);
+  rc |= run_dkv_case(2, 257, 64, true, false, false);
+  rc |= run_dkv_case(10240, 256, 256, true, false, true);
   printf(rc == 0 ? "ALL PASS\n" : "FAILURES PRESENT\n");
   return rc;
 }
