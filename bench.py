@@ -43,6 +43,11 @@ def parse_args():
                    help='enable per-block activation checkpointing '
                         '(needed only when activations exceed HBM; '
                         'costs a full forward recompute)')
+    p.add_argument('--checkpoint-ffs', action='store_true',
+                   help='selective FF-only checkpointing: frees the '
+                        '8x-dim GEGLU hiddens (the bulk of activation '
+                        'memory) at a 2-GEMM recompute cost — pushes '
+                        'per-GPU batch beyond 6')
     p.add_argument('--device', type=str, default=None)
     return p.parse_args()
 
@@ -76,7 +81,8 @@ def main():
         dim_head=args.dim_head,
         max_seq_len=max(2048, args.crop_len),
         reversible=args.reversible,
-        checkpoint_blocks=args.checkpoint_blocks,
+        checkpoint_blocks=('ff' if args.checkpoint_ffs
+                           else args.checkpoint_blocks),
     ).to(device)
     model.train()
 

@@ -406,3 +406,30 @@ def test_rotary_embeddings_functional():
     conv = DepthWiseConv1d(8, 16, kernel_size=3, padding=1)
     y = conv(torch.randn(2, 8, 12))
     assert y.shape == (2, 16, 12)
+
+
+def test_selective_ff_checkpointing_gradient_exact():
+    """checkpoint_blocks='ff' (recompute only the FF transitions) must
+    give bit-identical gradients to no checkpointing."""
+    torch.manual_seed(0)
+    ref = tiny_model(checkpoint_blocks=False)
+    torch.manual_seed(0)
+    sel = tiny_model(checkpoint_blocks='ff')
+    sel.load_state_dict(ref.state_dict())
+
+    seq = torch.randint(0, 21, (1, 16))
+    msa = torch.randint(0, 21, (1, 3, 16))
+
+    for model in (ref, sel):
+        model.train()
+        torch.manual_seed(7)  # identical MLM corruption
+        ret = model(seq, msa)
+        (ret.distance.pow(2).mean() + ret.msa_mlm_loss).backward()
+
+    for (n1, p1), (n2, p2) in zip(ref.named_parameters(),
+                                  sel.named_parameters()):
+        assert n1 == n2
+        if p1.grad is None:
+            assert p2.grad is None
+            continue
+        assert torch.equal(p1.grad, p2.grad), n1
