@@ -345,7 +345,7 @@ def test_overfit_single_batch_cpu():
             ret.distance.permute(0, 3, 1, 2), tgt, ignore_index=-100)
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] * 0.7, losses
 
 
