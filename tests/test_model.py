@@ -433,3 +433,45 @@ def test_selective_ff_checkpointing_gradient_exact():
             assert p2.grad is None
             continue
         assert torch.equal(p1.grad, p2.grad), n1
+
+
+def test_forward_kwarg_combinations():
+    """Combinatorial smoke over optional-input interactions: every
+    combination of {msa|embedds} x templates x angles x coords x mask
+    must run (the reference crashes on several of these)."""
+    import itertools
+    model = tiny_model(predict_angles=True, predict_coords=True,
+                       templates_dim=16, templates_angles_feats_dim=16,
+                       structure_module_depth=1).eval()
+    n, t = 12, 2
+    seq = torch.randint(0, 21, (1, n))
+    msa = torch.randint(0, 21, (1, 3, n))
+    embedds = torch.randn(1, 1, n, 1280)
+    tf = torch.randn(1, t, n, n, 16)
+    ta = torch.randn(1, t, n, 16)
+    tm = torch.ones(1, t, n).bool()
+
+    for (use_embedds, use_templates, use_angles_feats, use_mask,
+         return_trunk) in itertools.product([False, True], repeat=5):
+        kwargs = {}
+        if use_embedds:
+            args = (seq, None)
+            kwargs['embedds'] = embedds
+        else:
+            args = (seq, msa)
+        if use_mask:
+            kwargs['mask'] = torch.ones(1, n).bool()
+        if use_templates:
+            kwargs['templates_feats'] = tf
+            kwargs['templates_mask'] = tm
+            if use_angles_feats:
+                kwargs['templates_angles'] = ta
+        elif use_angles_feats:
+            continue  # angles require templates
+        kwargs['return_trunk'] = return_trunk
+        with torch.no_grad():
+            out = model(*args, **kwargs)
+        if return_trunk:
+            assert out.distance.shape == (1, n, n, 37)
+        else:
+            assert out.shape == (1, n, 3)
