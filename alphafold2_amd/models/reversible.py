@@ -38,7 +38,7 @@ class Deterministic(nn.Module):
 
     def record_rng(self, *args):
         self.cpu_state = torch.get_rng_state()
-        if torch.cuda._initialized:
+        if torch.cuda.is_initialized():
             self.cuda_in_fwd = True
             self.gpu_devices, self.gpu_states = get_device_states(*args)
 
