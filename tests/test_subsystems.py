@@ -300,6 +300,7 @@ def test_predict_cli_a3m(tmp_path):
     assert os.path.exists(out_pdb)
 
 
+@pytest.mark.filterwarnings('ignore::UserWarning')  # upstream kineto notice
 def test_profiling_helpers(tmp_path):
     from alphafold2_amd.runtime.profiling import (
         StepTimer, kernel_stats_summary, profile_trace)
