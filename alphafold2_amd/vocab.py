@@ -235,10 +235,12 @@ SC_BUILD = {
           (7, (1, 4, 5), 1.43, 126.6, -90.0),
           (8, (4, 5, 6), 1.38, 110.0, 180.0),
           (9, (4, 5, 7), 1.40, 107.0, 180.0),
-          (10, (5, 7, 9), 1.40, _TRI, 0.0),
-          (11, (7, 9, 8), 1.39, _TRI, 180.0),
-          (12, (9, 10, 11), 1.37, _TRI, 0.0),
-          (13, (10, 11, 12), 1.37, _TRI, 0.0)],
+          # benzene ring of the indole: each atom placed from its TRUE
+          # bond parent (c of the frame) with planar ring torsions
+          (10, (8, 9, 7), 1.40, _TRI, 180.0),   # CE3 on CD2
+          (11, (10, 7, 9), 1.40, _TRI, 0.0),    # CZ2 on CE2
+          (12, (9, 7, 10), 1.39, _TRI, 0.0),    # CZ3 on CE3
+          (13, (7, 10, 12), 1.37, _TRI, 0.0)],  # CH2 on CZ3
     'Y': [(4, (0, 2, 1), _CC, _TET, 122.7),
           (5, (0, 1, 4), _CC, _TET, 180.0),
           (6, (1, 4, 5), 1.39, _TRI, 90.0),

@@ -301,3 +301,38 @@ if HAS_HYP:
         assert buckets.min() >= 0 and buckets.max() <= 36
         # symmetry
         assert torch.equal(buckets, buckets.transpose(1, 2))
+
+
+def test_sidechain_ring_closure():
+    """Every chemically true side-chain bond — including the aromatic
+    ring-closure bonds of F/Y/H/W and proline's CD-N — must come out of
+    the NeRF builder at a plausible bond length (regression: the W
+    indole benzene ring was built from wrong parents)."""
+    from alphafold2_amd.vocab import SC_ATOM_NAMES
+    TRUE_BONDS = {
+        'F': [('CB','CG'),('CG','CD1'),('CG','CD2'),('CD1','CE1'),
+              ('CD2','CE2'),('CE1','CZ'),('CE2','CZ')],
+        'Y': [('CB','CG'),('CG','CD1'),('CG','CD2'),('CD1','CE1'),
+              ('CD2','CE2'),('CE1','CZ'),('CE2','CZ'),('CZ','OH')],
+        'H': [('CB','CG'),('CG','ND1'),('CG','CD2'),('ND1','CE1'),
+              ('CD2','NE2'),('CE1','NE2')],
+        'W': [('CB','CG'),('CG','CD1'),('CG','CD2'),('CD1','NE1'),
+              ('NE1','CE2'),('CD2','CE2'),('CD2','CE3'),('CE2','CZ2'),
+              ('CE3','CZ3'),('CZ2','CH2'),('CZ3','CH2')],
+        'P': [('CB','CG'),('CG','CD'),('CD','N')],
+    }
+    n = torch.tensor([0., 0., 0.])
+    ca = torch.tensor([1.46, 0., 0.])
+    c = torch.tensor([2.0, 1.42, 0.])
+    o = torch.tensor([1.6, 2.5, 0.])
+    res = torch.stack([n, ca, c, o])
+    from alphafold2_amd.vocab import VOCAB
+    for aa, bonds in TRUE_BONDS.items():
+        names = ['N', 'CA', 'C', 'O'] + SC_ATOM_NAMES[aa]
+        out = sidechain_container(
+            torch.tensor([[VOCAB._char2int[aa]]]), res[None],
+            atom_mask=torch.tensor([1] * 4 + [0] * 10))
+        coords = out[0, 0]
+        for a, b in bonds:
+            d = (coords[names.index(a)] - coords[names.index(b)]).norm()
+            assert 1.2 < d.item() < 1.95, (aa, a, b, d.item())
