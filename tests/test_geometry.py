@@ -336,3 +336,21 @@ def test_sidechain_ring_closure():
         for a, b in bonds:
             d = (coords[names.index(a)] - coords[names.index(b)]).norm()
             assert 1.2 < d.item() < 1.95, (aa, a, b, d.item())
+
+
+def test_sidechain_no_intra_residue_clashes():
+    # no two atoms of any built residue may overlap (< 1.15 A)
+    from alphafold2_amd.vocab import VOCAB, SC_ATOM_NAMES
+    n = torch.tensor([0., 0., 0.])
+    ca = torch.tensor([1.46, 0., 0.])
+    c = torch.tensor([2.0, 1.42, 0.])
+    o = torch.tensor([1.6, 2.5, 0.])
+    res = torch.stack([n, ca, c, o])
+    for aa_id in range(20):
+        aa = VOCAB._int2char[aa_id]
+        nat = 4 + len(SC_ATOM_NAMES[aa])
+        out = sidechain_container(torch.tensor([[aa_id]]), res[None],
+                                  atom_mask=torch.tensor([1] * 4 + [0] * 10))
+        coords = out[0, 0, :nat]
+        d = torch.cdist(coords, coords) + torch.eye(nat) * 99
+        assert d.min().item() > 1.15, (aa, d.min().item())
