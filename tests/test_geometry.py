@@ -354,3 +354,27 @@ def test_sidechain_no_intra_residue_clashes():
         coords = out[0, 0, :nat]
         d = torch.cdist(coords, coords) + torch.eye(nat) * 99
         assert d.min().item() > 1.15, (aa, d.min().item())
+
+
+def test_mirror_fix_enforces_protein_chirality():
+    # MDS reconstruction is chirality-blind; fix_mirror must leave the
+    # output with a negative-phi MAJORITY (the biological convention)
+    # whichever mirror image the distances came from.
+    import math
+    from alphafold2_amd.geometry.mds import calc_phis_torch
+    L = 24
+    pts = []
+    for i in range(L * 3):
+        t = i * (2 * math.pi / 10.8)
+        pts.append([2.3 * math.cos(t), 2.3 * math.sin(t), 0.5 * i])
+    coords = torch.tensor(pts)              # (N, 3) helix
+    for flip in (1.0, -1.0):                # both mirror images
+        c = coords.clone()
+        c[:, 2] *= flip
+        dist = torch.cdist(c[None], c[None])[0]
+        masker = torch.arange(L * 3) % 3
+        N_mask, CA_mask = (masker == 0), (masker == 1)
+        out, _ = MDScaling(dist, iters=60, fix_mirror=True,
+                           N_mask=N_mask, CA_mask=CA_mask, C_mask=None)
+        frac = calc_phis_torch(out, N_mask, CA_mask, prop=True)[0]
+        assert frac >= 0.5, (flip, float(frac))
