@@ -156,6 +156,15 @@ def test_fake_embedder_wrappers():
                        msa_mask=msa_mask)
     assert ret.distance.shape == (2, 16, 16, 37)
 
+    # ragged MSA depth: fully-padded rows must be excluded from the
+    # row-tied embedding then re-padded (per-batch-element branch)
+    ragged = msa_mask.clone()
+    ragged[0, -1] = False
+    with torch.no_grad():
+        ret = wrapper2(seq, msa, mask=torch.ones_like(seq).bool(),
+                       msa_mask=ragged)
+    assert ret.distance.shape == (2, 16, 16, 37)
+
 
 def test_mlm_noise_and_loss():
     mlm = MLM(dim=16, num_tokens=21, mask_id=21)
