@@ -52,7 +52,7 @@ def _ddp_worker(rank, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(120)
+@pytest.mark.timeout(300)
 def test_ddp_grad_allreduce():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
@@ -66,7 +66,7 @@ def test_ddp_grad_allreduce():
         rank, psum, grads = q.get()
         results[rank] = (psum, grads)
     for p in procs:
-        p.join(timeout=60)
+        p.join(timeout=240)
         assert p.exitcode == 0
 
     # parameter broadcast: identical initial weights
@@ -112,7 +112,7 @@ def _no_sync_worker(rank, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(120)
+@pytest.mark.timeout(300)
 def test_ddp_no_sync_accumulation():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
@@ -126,7 +126,7 @@ def test_ddp_no_sync_accumulation():
         rank, grads = q.get()
         results[rank] = grads
     for p in procs:
-        p.join(timeout=60)
+        p.join(timeout=240)
         assert p.exitcode == 0
     assert torch.allclose(results[0], results[1], atol=1e-6)
 
@@ -166,7 +166,7 @@ def _unused_param_worker(rank, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(120)
+@pytest.mark.timeout(300)
 def test_ddp_unused_param_in_sync_micro():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
@@ -180,7 +180,7 @@ def test_ddp_unused_param_in_sync_micro():
         rank, x1, x2, grads = q.get()
         results[rank] = (x1, x2, grads)
     for p in procs:
-        p.join(timeout=60)
+        p.join(timeout=240)
         assert p.exitcode == 0
 
     assert torch.allclose(results[0][2], results[1][2], atol=1e-6)
@@ -246,7 +246,7 @@ def test_ddp_full_model_step():
         rank, gsum = q.get()
         results[rank] = gsum
     for p in procs:
-        p.join(timeout=120)
+        p.join(timeout=240)
         assert p.exitcode == 0
     # averaged grads identical on both ranks
     g0 = torch.as_tensor(results[0])
@@ -280,7 +280,7 @@ def _never_used_worker(rank, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(120)
+@pytest.mark.timeout(300)
 def test_ddp_never_used_bucket_skipped():
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
@@ -294,7 +294,7 @@ def test_ddp_never_used_bucket_skipped():
         rank, b_skipped, a_used, a_grad = q.get()
         results[rank] = (b_skipped, a_used, a_grad)
     for p in procs:
-        p.join(timeout=60)
+        p.join(timeout=240)
         assert p.exitcode == 0
     for r in range(WORLD):
         assert results[r][0], 'unused bucket must never allocate/reduce'
