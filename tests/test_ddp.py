@@ -47,7 +47,7 @@ def _ddp_worker(rank, port, q):
     engine.finalize()
 
     grads = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
-    q.put((rank, psum, grads))
+    q.put((rank, psum, grads.numpy().copy()))
     dist.barrier()
     dist.destroy_process_group()
 
@@ -73,7 +73,8 @@ def test_ddp_grad_allreduce():
     assert abs(results[0][0] - results[1][0]) < 1e-6
 
     # both ranks end with identical (averaged) gradients
-    g0, g1 = results[0][1], results[1][1]
+    g0 = torch.as_tensor(results[0][1])
+    g1 = torch.as_tensor(results[1][1])
     assert torch.allclose(g0, g1, atol=1e-6)
 
     # and they equal the single-process average of per-rank gradients
@@ -107,7 +108,7 @@ def _no_sync_worker(rank, port, q):
     model(x2).pow(2).sum().backward()
     engine.finalize()
     grads = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
-    q.put((rank, grads))
+    q.put((rank, grads.numpy().copy()))
     dist.barrier()
     dist.destroy_process_group()
 
@@ -128,7 +129,8 @@ def test_ddp_no_sync_accumulation():
     for p in procs:
         p.join(timeout=240)
         assert p.exitcode == 0
-    assert torch.allclose(results[0], results[1], atol=1e-6)
+    assert torch.allclose(torch.as_tensor(results[0]),
+                          torch.as_tensor(results[1]), atol=1e-6)
 
 
 class _TwoPath(torch.nn.Module):
@@ -161,7 +163,7 @@ def _unused_param_worker(rank, port, q):
     model(x2, use_b=False).pow(2).sum().backward()  # b unused here
     engine.finalize()
     grads = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
-    q.put((rank, x1, x2, grads))
+    q.put((rank, x1.numpy(), x2.numpy(), grads.numpy().copy()))
     dist.barrier()
     dist.destroy_process_group()
 
@@ -183,7 +185,8 @@ def test_ddp_unused_param_in_sync_micro():
         p.join(timeout=240)
         assert p.exitcode == 0
 
-    assert torch.allclose(results[0][2], results[1][2], atol=1e-6)
+    assert torch.allclose(torch.as_tensor(results[0][2]),
+                          torch.as_tensor(results[1][2]), atol=1e-6)
 
     # equals the all-rank average of locally accumulated gradients
     torch.manual_seed(5)
@@ -191,14 +194,16 @@ def test_ddp_unused_param_in_sync_micro():
     expected = None
     for rank in range(WORLD):
         model.zero_grad()
-        x1, x2 = results[rank][0], results[rank][1]
+        x1 = torch.as_tensor(results[rank][0])
+        x2 = torch.as_tensor(results[rank][1])
         model(x1, use_b=True).pow(2).sum().backward()
         model(x2, use_b=False).pow(2).sum().backward()
         g = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
         expected = g if expected is None else expected + g
     expected = expected / WORLD
-    assert torch.allclose(results[0][2], expected, atol=1e-5), \
-        (results[0][2] - expected).abs().max()
+    g0 = torch.as_tensor(results[0][2])
+    assert torch.allclose(g0, expected, atol=1e-5), \
+        (g0 - expected).abs().max()
 
 
 def _model_worker(rank, port, q):
