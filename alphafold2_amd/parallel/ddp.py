@@ -71,7 +71,7 @@ def all_reduce_mean(t: torch.Tensor):
 
 class _Bucket:
     __slots__ = ('params', 'flat', 'numel', 'ready', 'work', 'offsets',
-                 'seen', 'ever_used')
+                 'seen', 'pending', 'skipped')
 
     def __init__(self):
         self.params: List[torch.nn.Parameter] = []
@@ -81,12 +81,11 @@ class _Bucket:
         self.work = None
         self.offsets = {}
         self.seen = set()
-        # becomes True the first time ANY param in the bucket produces a
-        # gradient; never-used buckets (e.g. template modules in a
-        # template-less run) skip their all-reduce entirely.  Monotonic,
-        # so ranks stay collective-consistent under the standard DDP
-        # contract (all ranks execute the same graph).
-        self.ever_used = False
+        # all params got grads this step; the bucket is eligible to
+        # launch (launches happen strictly in bucket-index order)
+        self.pending = False
+        # skip_unused_buckets marked this bucket as reduce-less this step
+        self.skipped = False
 
 
 class DataParallelEngine:
@@ -102,16 +101,30 @@ class DataParallelEngine:
     """
 
     def __init__(self, model: torch.nn.Module, bucket_cap_mb: float = 64,
-                 process_group=None, grad_dtype=None):
+                 process_group=None, grad_dtype=None,
+                 skip_unused_buckets: bool = False):
         self.model = model
         self.group = process_group
         self.world_size = get_world_size()
         self.enabled = is_distributed()
         self.grad_dtype = grad_dtype
+        # skip_unused_buckets=True suppresses the all-reduce of buckets
+        # in which NO param has a grad.  Only safe when every rank runs
+        # the SAME graph every step (optional inputs like templates /
+        # extra-MSA present-or-absent uniformly across ranks): a rank
+        # skipping a bucket that another rank reduces deadlocks the job.
+        # Default False = always reduce every bucket (collective-
+        # consistent by construction, like torch DDP).
+        self.skip_unused_buckets = skip_unused_buckets
         self._hooks = []
         self._buckets: List[_Bucket] = []
         self._param_bucket = {}
         self._sync = True
+        # index of the next bucket allowed to launch its all-reduce.
+        # Buckets launch strictly in index order regardless of the order
+        # their hooks complete, so ranks whose backward visits modules
+        # in different orders still issue identical collective sequences.
+        self._next_launch = 0
 
         if self.enabled:
             self._broadcast_parameters()
@@ -179,34 +192,46 @@ class DataParallelEngine:
         off = bucket.offsets[p]
         bucket.flat[off:off + p.numel()].copy_(p.grad.reshape(-1))
         bucket.seen.add(p)
-        bucket.ever_used = True
         bucket.ready += 1
         if bucket.ready == len(bucket.params):
-            # one large message per bucket; async so backward keeps going
-            bucket.work = dist.all_reduce(bucket.flat, op=dist.ReduceOp.SUM,
-                                          group=self.group, async_op=True)
+            bucket.pending = True
+            self._launch_in_order()
+
+    def _launch_in_order(self):
+        """Launch pending bucket all-reduces strictly in bucket-index
+        order.  One large message per bucket; async so backward keeps
+        going while early buckets reduce."""
+        while self._next_launch < len(self._buckets) \
+                and self._buckets[self._next_launch].pending:
+            bucket = self._buckets[self._next_launch]
+            if not bucket.skipped:
+                bucket.work = dist.all_reduce(
+                    bucket.flat, op=dist.ReduceOp.SUM,
+                    group=self.group, async_op=True)
+            self._next_launch += 1
 
     def finalize(self):
-        """Wait for in-flight collectives, average, write grads back."""
+        """Flush buckets the hooks didn't complete, wait for in-flight
+        collectives, average, write grads back."""
         if not self.enabled:
             return
         inv = 1.0 / self.world_size
+        # fill + mark every not-yet-pending bucket, then launch the
+        # remainder in index order (identical sequence on every rank)
         for bucket in self._buckets:
-            if not bucket.ever_used:
-                # no param in this bucket has EVER produced a gradient
-                # (this step included) -> its contribution is zero on
-                # every rank; skip the collective
-                if not any(p.grad is not None for p in bucket.params):
-                    continue
-                bucket.ever_used = True
-            if bucket.ready != len(bucket.params):
-                # params that never got grads this step (unused path):
-                # reduce what we have for deterministic behavior
-                self._flush_partial(bucket)
+            if bucket.pending:
+                continue
+            if self.skip_unused_buckets \
+                    and not any(p.grad is not None for p in bucket.params):
+                bucket.skipped = True
+            else:
+                self._fill_flat(bucket)
+            bucket.pending = True
+        self._launch_in_order()
+        for bucket in self._buckets:
             if bucket.work is not None:
                 bucket.work.wait()
                 bucket.work = None
-            if bucket.flat is not None:
                 bucket.flat.mul_(inv)
                 for p in bucket.params:
                     if p.grad is None:
@@ -215,23 +240,25 @@ class DataParallelEngine:
                     p.grad.copy_(
                         bucket.flat[off:off + p.numel()].view_as(p.grad))
             bucket.ready = 0
+            bucket.pending = False
+            bucket.skipped = False
             bucket.seen.clear()
+        self._next_launch = 0
 
-    def _flush_partial(self, bucket: _Bucket):
-        """Reduce a bucket some of whose params never fired their hook
-        this step.  A param can hold a REAL accumulated grad and still
-        not fire (it got its grad during a no_sync micro-batch and was
-        unused in the final one), so copy every un-seen grad into the
-        flat buffer — and zero the slice of truly grad-less params so a
-        previous step's values never leak into the collective."""
-        device = next(self.model.parameters()).device
-        any_grad = next((p.grad for p in bucket.params if p.grad is not None),
-                        None)
-        if any_grad is None and bucket.flat is None:
-            self._ensure_flat(
-                bucket, torch.zeros(1, device=device))
-        elif bucket.flat is None:
-            self._ensure_flat(bucket, any_grad)
+    def _fill_flat(self, bucket: _Bucket):
+        """Fill the flat buffer of a bucket some of whose params never
+        fired their hook this step.  A param can hold a REAL accumulated
+        grad and still not fire (it got its grad during a no_sync
+        micro-batch and was unused in the final one), so copy every
+        un-seen grad into the flat buffer — and zero the slice of truly
+        grad-less params so a previous step's values never leak into the
+        collective."""
+        if bucket.flat is None:
+            any_grad = next(
+                (p.grad for p in bucket.params if p.grad is not None), None)
+            like = any_grad if any_grad is not None \
+                else next(self.model.parameters())
+            self._ensure_flat(bucket, like)
         for p in bucket.params:
             if p in bucket.seen:
                 continue
@@ -241,8 +268,6 @@ class DataParallelEngine:
                 dst.copy_(p.grad.reshape(-1))
             else:
                 dst.zero_()
-        bucket.work = dist.all_reduce(bucket.flat, op=dist.ReduceOp.SUM,
-                                      group=self.group, async_op=True)
         bucket.ready = len(bucket.params)
 
     def zero_grad(self, set_to_none: bool = True):
@@ -250,7 +275,10 @@ class DataParallelEngine:
         for bucket in self._buckets:
             bucket.ready = 0
             bucket.work = None
+            bucket.pending = False
+            bucket.skipped = False
             bucket.seen.clear()
+        self._next_launch = 0
 
     def remove(self):
         for h in self._hooks:

@@ -261,13 +261,15 @@ def test_ddp_full_model_step():
 
 
 def _never_used_worker(rank, port, q):
-    """A bucket whose params never produce grads must skip its
-    all-reduce (flat never allocated) while used buckets still average."""
+    """With skip_unused_buckets=True (uniform-graph contract), a bucket
+    whose params never produce grads must skip its all-reduce (flat
+    never allocated) while used buckets still average."""
     from alphafold2_amd.parallel import DataParallelEngine
     _setup(rank, WORLD, port)
     torch.manual_seed(3)
     model = _TwoPath()
-    engine = DataParallelEngine(model, bucket_cap_mb=0.0001)  # per-param
+    engine = DataParallelEngine(model, bucket_cap_mb=0.0001,  # per-param
+                                skip_unused_buckets=True)
 
     torch.manual_seed(rank)
     x = torch.randn(2, 8)
@@ -276,10 +278,8 @@ def _never_used_worker(rank, port, q):
         model(x, use_b=False).pow(2).sum().backward()
         engine.finalize()
     b_buckets = [engine._param_bucket[p] for p in model.b.parameters()]
-    a_buckets = [engine._param_bucket[p] for p in model.a.parameters()]
     q.put((rank,
-           all(bk.flat is None and not bk.ever_used for bk in b_buckets),
-           all(bk.ever_used for bk in a_buckets),
+           all(bk.flat is None for bk in b_buckets),
            model.a.weight.grad.numpy().copy()))
     dist.barrier()
     dist.destroy_process_group()
@@ -296,13 +296,72 @@ def test_ddp_never_used_bucket_skipped():
         p.start()
     results = {}
     for _ in range(WORLD):
-        rank, b_skipped, a_used, a_grad = q.get()
-        results[rank] = (b_skipped, a_used, a_grad)
+        rank, b_skipped, a_grad = q.get()
+        results[rank] = (b_skipped, a_grad)
     for p in procs:
         p.join(timeout=240)
         assert p.exitcode == 0
     for r in range(WORLD):
         assert results[r][0], 'unused bucket must never allocate/reduce'
-        assert results[r][1]
-    assert torch.allclose(torch.as_tensor(results[0][2]),
-                          torch.as_tensor(results[1][2]), atol=1e-6)
+    assert torch.allclose(torch.as_tensor(results[0][1]),
+                          torch.as_tensor(results[1][1]), atol=1e-6)
+
+
+def _divergent_graph_worker(rank, port, q):
+    """Ranks running DIFFERENT graphs in the same step (rank 0 uses the
+    `b` branch, rank 1 does not — the realistic per-rank-data case from
+    the round-1 advisory).  Default engine settings must neither hang
+    nor mix buckets: the non-user contributes zeros and both ranks end
+    with the same average."""
+    from alphafold2_amd.parallel import DataParallelEngine
+    _setup(rank, WORLD, port)
+    torch.manual_seed(11)
+    model = _TwoPath()
+    engine = DataParallelEngine(model, bucket_cap_mb=0.0001)  # per-param
+
+    torch.manual_seed(rank * 17)
+    x = torch.randn(2, 8)
+    model(x, use_b=(rank == 0)).pow(2).sum().backward()
+    engine.finalize()
+    grads = {n: (p.grad.numpy().copy() if p.grad is not None else None)
+             for n, p in model.named_parameters()}
+    q.put((rank, x.numpy(), grads))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_divergent_graphs_across_ranks():
+    ctx = mp.get_context('spawn')
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_divergent_graph_worker, args=(r, port, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, x, grads = q.get()
+        results[rank] = (x, grads)
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+
+    # expected: average over ranks, zeros where a rank skipped the branch
+    torch.manual_seed(11)
+    model = _TwoPath()
+    expected = {n: torch.zeros_like(p) for n, p in model.named_parameters()}
+    for rank in range(WORLD):
+        model.zero_grad()
+        x = torch.as_tensor(results[rank][0])
+        model(x, use_b=(rank == 0)).pow(2).sum().backward()
+        for n, p in model.named_parameters():
+            if p.grad is not None:
+                expected[n] += p.grad / WORLD
+
+    for rank in range(WORLD):
+        for n, g in results[rank][1].items():
+            if g is None:
+                continue  # rank never produced a local grad for n
+            assert torch.allclose(torch.as_tensor(g), expected[n],
+                                  atol=1e-6), (rank, n)
