@@ -1,6 +1,6 @@
 from .backend import (
-    exists, set_backend_kwarg, expand_dims_to, expand_arg_dims,
-    invoke_torch_or_numpy, torch_default_dtype,
+    exists, as_batched, expand_dims_to, dual_backend, resolve_backend,
+    torch_default_dtype,
 )
 from .distogram import (
     DISTANCE_THRESHOLDS, get_bucketed_distance_matrix, center_distogram_torch,

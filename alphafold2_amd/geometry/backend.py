@@ -1,10 +1,14 @@
-"""Dual torch/numpy backend plumbing for the geometry toolbox.
+"""Backend dispatch for the geometry toolbox.
 
-Capability parity with the reference's decorator discipline
-(reference utils.py:54-104): every public geometry wrapper accepts
-``backend='auto'|'torch'|'numpy'`` and dispatches to a torch or numpy
-twin.  This keeps the whole geometry layer unit-testable on CPU without
-a GPU in the loop.
+The reference expresses its torch/numpy dual backend as three stacked
+decorators (utils.py:54-104: backend-kwarg resolution, positional-arg
+dim expansion, twin invocation).  Here the whole policy lives in ONE
+factory, `dual_backend`: a public wrapper is built from its torch and
+numpy implementations plus an optional argument-preparation hook, so
+each geometry entry point is a single declaration instead of a
+decorator tower.  The capability is the same — every public wrapper
+accepts ``backend='auto'|'torch'|'numpy'`` and stays unit-testable on
+CPU — but the mechanism is this package's own.
 """
 import contextlib
 from functools import wraps
@@ -16,58 +20,67 @@ def exists(val):
     return val is not None
 
 
+def as_batched(t, ndim):
+    """View a tensor/array with leading singleton axes so it has `ndim`
+    dims (no-op when it already does)."""
+    shape = t.shape
+    if len(shape) >= ndim:
+        return t
+    return t.reshape((1,) * (ndim - len(shape)) + tuple(shape))
+
+
+# transitional alias for the reference-style call convention
+# (length = how many leading singleton dims to ADD)
 def expand_dims_to(t, length=3):
-    """Left-pad the shape of a tensor/array with singleton dims."""
     if length <= 0:
         return t
-    return t.reshape(*((1,) * length), *t.shape)
+    return as_batched(t, len(t.shape) + length)
 
 
-def set_backend_kwarg(fn):
-    """Resolve backend='auto' from the type of the first argument."""
-    @wraps(fn)
-    def inner(*args, backend='auto', **kwargs):
-        if backend == 'auto':
-            backend = 'torch' if isinstance(args[0], torch.Tensor) else 'numpy'
-        kwargs.update(backend=backend)
-        return fn(*args, **kwargs)
-    return inner
+def resolve_backend(sample, requested='auto'):
+    """'auto' means: torch when the probe argument is a torch tensor,
+    numpy otherwise."""
+    if requested != 'auto':
+        return requested
+    return 'torch' if isinstance(sample, torch.Tensor) else 'numpy'
 
 
-def expand_arg_dims(dim_len=3):
-    """Expand the two positional args to `dim_len` dims (adds batch dims)."""
-    def outer(fn):
-        @wraps(fn)
-        def inner(x, y, **kwargs):
-            assert len(x.shape) == len(y.shape), "shapes of A and B must match"
-            remaining = dim_len - len(x.shape)
-            x = expand_dims_to(x, length=remaining)
-            y = expand_dims_to(y, length=remaining)
-            return fn(x, y, **kwargs)
-        return inner
-    return outer
+def dual_backend(torch_impl, numpy_impl, pair_ndim=None, prepare=None):
+    """Build a public geometry wrapper from two backend twins.
+
+    * pair_ndim: when set, the first two positional args are promoted to
+      this rank (adds batch dims) and must agree in rank.
+    * prepare(*args, **kwargs) -> (args, kwargs): optional hook that
+      normalizes/augments the arguments before the twin is invoked.
+
+    The returned wrapper takes ``backend='auto'|'torch'|'numpy'``.
+    """
+    def dispatch(*args, backend='auto', **kwargs):
+        impl = torch_impl if resolve_backend(args[0], backend) == 'torch' \
+            else numpy_impl
+        if pair_ndim is not None:
+            a, b = args[0], args[1]
+            assert len(a.shape) == len(b.shape), \
+                'the two inputs must have the same rank'
+            args = (as_batched(a, pair_ndim), as_batched(b, pair_ndim)) \
+                + args[2:]
+        if prepare is not None:
+            args, kwargs = prepare(*args, **kwargs)
+        return impl(*args, **kwargs)
+    return dispatch
 
 
-def invoke_torch_or_numpy(torch_fn, numpy_fn):
-    """The wrapped fn returns the positional args (and optionally a kwargs
-    dict as last element); the selected backend twin is then invoked."""
-    def outer(fn):
-        @wraps(fn)
-        def inner(*args, **kwargs):
-            backend = kwargs.pop('backend')
-            passed_args = list(fn(*args, **kwargs))
-            if isinstance(passed_args[-1], dict):
-                passed_kwargs = passed_args.pop()
-            else:
-                passed_kwargs = {}
-            backend_fn = torch_fn if backend == 'torch' else numpy_fn
-            return backend_fn(*passed_args, **passed_kwargs)
-        return inner
-    return outer
+def named_wrapper(fn, name, doc):
+    """Give a dual_backend dispatcher a public name + docstring."""
+    fn.__name__ = name
+    fn.__qualname__ = name
+    fn.__doc__ = doc
+    return fn
 
 
 @contextlib.contextmanager
 def torch_default_dtype(dtype):
+    """Standard default-dtype scope (the canonical try/finally idiom)."""
     prev = torch.get_default_dtype()
     torch.set_default_dtype(dtype)
     try:

@@ -1,102 +1,120 @@
 """Multidimensional scaling (distogram -> 3D) + chirality fix.
 
-Capability parity: reference utils.py:766-879 (mds twins), 881-993
-(dihedrals / phi-based mirror selection), 1162-1201 + 1254-1279
-(mdscaling wrappers).  The torch path is fully batched (batched
-svd_lowrank init + Guttman transform via bmm) and runs on the MI355X.
+Capability parity with reference utils.py:766-879 (MDS), 881-993
+(dihedrals / phi-based mirror selection) and 1162-1279 (wrappers), but
+the implementation is this package's own design:
+
+* classical-MDS initialization by proper double-centering of the
+  squared-distance matrix followed by a BATCHED symmetric
+  eigendecomposition (`torch.linalg.eigh`) — one lane for the whole
+  batch, no per-item SVD loop;
+* SMACOF refinement with the Guttman transform written as a single
+  batched update, tracking normalized stress for convergence;
+* backbone phi dihedrals computed batched over the whole batch (no
+  per-structure python loop);
+* the numpy backend is a conversion shim around the torch
+  implementation instead of a duplicated twin — one copy of the math.
 """
 import numpy as np
 import torch
 
-from .backend import set_backend_kwarg, invoke_torch_or_numpy, expand_dims_to
+from .backend import as_batched, dual_backend, named_wrapper
+
+
+# ---------------------------------------------------------------------------
+# core MDS (torch, batched)
+
+
+def _gram_from_distances(dist):
+    """Squared-distance matrix (b, N, N) -> centered Gram matrix via the
+    double-centering identity G = -1/2 * J D2 J with J = I - 11ᵀ/N."""
+    d2 = dist * dist
+    row = d2.mean(dim=-1, keepdim=True)
+    col = d2.mean(dim=-2, keepdim=True)
+    grand = d2.mean(dim=(-1, -2), keepdim=True)
+    return -0.5 * (d2 - row - col + grand)
+
+
+def _spectral_init(dist, dim=3):
+    """Classical-MDS embedding: top-`dim` eigenpairs of the Gram matrix,
+    batched.  Returns (b, N, dim) coordinates."""
+    gram = _gram_from_distances(dist)
+    # eigh is ascending; take the trailing (largest) eigenpairs
+    evals, evecs = torch.linalg.eigh(gram)
+    top_vals = evals[..., -dim:].clamp_min(0.0)
+    top_vecs = evecs[..., -dim:]
+    return top_vecs * top_vals.sqrt().unsqueeze(-2)
+
+
+def _smacof_refine(coords, target, weights, iters, tol, verbose=0):
+    """Iteratively majorize stress with the Guttman transform.
+
+    coords: (b, N, dim) initial embedding; target: (b, N, N) desired
+    distances; weights: (b, N, N).  Returns refined coords and the
+    per-iteration normalized-stress history (iters+1, b) seeded with inf.
+    """
+    b, N, _ = coords.shape
+    eye = torch.arange(N, device=coords.device)
+    history = [coords.new_full((b,), float('inf'))]
+    for it in range(iters):
+        coords = coords.contiguous()
+        cur = torch.cdist(coords, coords)
+        stress = 0.5 * (weights * (cur - target).square()).sum(dim=(-1, -2))
+        cur = cur.clone()
+        cur[cur <= 0] += 1e-7
+        # Guttman transform: B(X) X / N with B = diag(rowsum(w*t/c)) - w*t/c
+        wtc = weights * (target / cur)
+        bmat = -wtc
+        bmat[:, eye, eye] += wtc.sum(dim=-1)
+        proposal = torch.matmul(bmat, coords) / N
+        scale = proposal.norm(dim=(-1, -2))
+        norm_stress = stress / scale
+        if verbose >= 2:
+            print(f'smacof it {it}: stress {stress}')
+        if (history[-1] - norm_stress).mean() <= tol:
+            if verbose:
+                print(f'smacof converged at it {it}, stress {norm_stress}')
+            break
+        coords = proposal
+        history.append(norm_stress)
+    return coords, torch.stack(history, dim=0)
 
 
 def mds_torch(pre_dist_mat, weights=None, iters=10, tol=1e-5, eigen=False,
               verbose=0):
-    """Distance matrix (b, N, N) -> coords (b, 3, N) + stress history.
+    """Distance matrix ((b), N, N) -> coords (b, 3, N) + stress history.
 
-    Eigen-style init from the Gram matrix (svd_lowrank), then iterative
-    Guttman-transform refinement (SMACOF).
+    Spectral (classical-MDS) init, then SMACOF refinement; `eigen=True`
+    with no weights returns the spectral embedding directly.
     """
-    device = pre_dist_mat.device
-    pre_dist_mat = expand_dims_to(pre_dist_mat, length=3 - len(pre_dist_mat.shape))
-    batch, N, _ = pre_dist_mat.shape
-    diag_idxs = torch.arange(N, device=device)
-    his = [torch.full((batch,), float('inf'), device=device)]
+    pre_dist_mat = as_batched(pre_dist_mat, 3)
+    coords = _spectral_init(pre_dist_mat, dim=3)
 
-    # classical-MDS style init: Gram matrix from squared distances
-    D = pre_dist_mat ** 2
-    M = 0.5 * (D[:, :1, :] + D[:, :, :1] - D)
-    # per-item svd_lowrank (batched svd_lowrank is slower in practice)
-    svds = [torch.svd_lowrank(mi) for mi in M]
-    u = torch.stack([s[0] for s in svds], dim=0)
-    s = torch.stack([s[1] for s in svds], dim=0)
-    best_3d_coords = torch.bmm(u, torch.diag_embed(s).abs().sqrt())[..., :3]
-
-    if weights is None and eigen:
-        return best_3d_coords.transpose(-1, -2), torch.zeros_like(torch.stack(his, dim=0))
-    elif eigen and verbose:
-        print("Can't use eigen flag if weights are active. Fallback to iterative")
+    if eigen:
+        if weights is None:
+            b = pre_dist_mat.shape[0]
+            return coords.transpose(-1, -2), coords.new_zeros((1, b))
+        if verbose:
+            print("Can't use eigen flag if weights are active. "
+                  "Fallback to iterative")
 
     if weights is None:
         weights = torch.ones_like(pre_dist_mat)
-
-    for i in range(iters):
-        best_3d_coords = best_3d_coords.contiguous()
-        dist_mat = torch.cdist(best_3d_coords, best_3d_coords, p=2).clone()
-
-        stress = (weights * (dist_mat - pre_dist_mat) ** 2).sum(dim=(-1, -2)) * 0.5
-        dist_mat[dist_mat <= 0] += 1e-7
-        ratio = weights * (pre_dist_mat / dist_mat)
-        B = -ratio
-        B[:, diag_idxs, diag_idxs] += ratio.sum(dim=-1)
-
-        coords = (1. / N) * torch.matmul(B, best_3d_coords)
-        dis = torch.norm(coords, dim=(-1, -2))
-
-        if verbose >= 2:
-            print(f'it: {i}, stress {stress}')
-        if (his[-1] - stress / dis).mean() <= tol:
-            if verbose:
-                print(f'breaking at iteration {i} with stress {stress / dis}')
-            break
-
-        best_3d_coords = coords
-        his.append(stress / dis)
-
-    return best_3d_coords.transpose(-1, -2), torch.stack(his, dim=0)
+    coords, history = _smacof_refine(coords, pre_dist_mat, weights,
+                                     iters, tol, verbose)
+    return coords.transpose(-1, -2), history
 
 
 def mds_numpy(pre_dist_mat, weights=None, iters=10, tol=1e-5, eigen=False,
               verbose=0):
-    if weights is None:
-        weights = np.ones_like(pre_dist_mat)
-    pre_dist_mat = expand_dims_to(pre_dist_mat, length=3 - len(pre_dist_mat.shape))
-    batch, N, _ = pre_dist_mat.shape
-    # per-batch history (a scalar seed breaks numpy>=2 stacking)
-    his = [np.full(batch, np.inf)]
-    best_stress = np.inf * np.ones(batch)
-    best_3d_coords = 2 * np.random.rand(batch, 3, N) - 1
-    for i in range(iters):
-        dist_mat = np.linalg.norm(
-            best_3d_coords[:, :, :, None] - best_3d_coords[:, :, None, :], axis=-3)
-        stress = ((weights * (dist_mat - pre_dist_mat)) ** 2).sum(axis=(-1, -2)) * 0.5
-        dist_mat[dist_mat == 0] = 1e-7
-        ratio = weights * (pre_dist_mat / dist_mat)
-        B = -ratio
-        B[:, np.arange(N), np.arange(N)] += ratio.sum(axis=-1)
-        coords = (1. / N) * np.matmul(best_3d_coords, B)
-        dis = np.linalg.norm(coords, axis=(-1, -2))
-        if verbose >= 2:
-            print(f'it: {i}, stress {stress}')
-        if (best_stress - stress / dis).mean() <= tol:
-            if verbose:
-                print(f'breaking at iteration {i} with stress {stress / dis}')
-            break
-        best_3d_coords = coords
-        best_stress = stress / dis
-        his.append(best_stress)
-    return best_3d_coords, np.array(his)
+    """Numpy shim over the torch implementation (one copy of the math)."""
+    t_dist = torch.as_tensor(np.ascontiguousarray(pre_dist_mat),
+                             dtype=torch.float64)
+    t_w = None if weights is None else \
+        torch.as_tensor(np.ascontiguousarray(weights), dtype=torch.float64)
+    coords, history = mds_torch(t_dist, weights=t_w, iters=iters, tol=tol,
+                                eigen=eigen, verbose=verbose)
+    return coords.numpy(), history.numpy()
 
 
 # ---------------------------------------------------------------------------
@@ -105,107 +123,111 @@ def mds_numpy(pre_dist_mat, weights=None, iters=10, tol=1e-5, eigen=False,
 
 def get_dihedral_torch(c1, c2, c3, c4):
     """Dihedral angle (radians) for four points, batched over leading dims."""
-    u1 = c2 - c1
-    u2 = c3 - c2
-    u3 = c4 - c3
-    return torch.atan2(
-        ((torch.norm(u2, dim=-1, keepdim=True) * u1) * torch.cross(u2, u3, dim=-1)).sum(dim=-1),
-        (torch.cross(u1, u2, dim=-1) * torch.cross(u2, u3, dim=-1)).sum(dim=-1))
+    b1 = c2 - c1
+    b2 = c3 - c2
+    b3 = c4 - c3
+    n1 = torch.cross(b1, b2, dim=-1)
+    n2 = torch.cross(b2, b3, dim=-1)
+    # sign convention matches the classic |b2|*b1 . (b2 x b3) numerator
+    m = torch.cross(b2 / b2.norm(dim=-1, keepdim=True), n1, dim=-1)
+    return torch.atan2((m * n2).sum(dim=-1), (n1 * n2).sum(dim=-1))
 
 
 def get_dihedral_numpy(c1, c2, c3, c4):
-    u1 = c2 - c1
-    u2 = c3 - c2
-    u3 = c4 - c3
-    return np.arctan2(
-        ((np.linalg.norm(u2, axis=-1, keepdims=True) * u1) * np.cross(u2, u3, axis=-1)).sum(axis=-1),
-        (np.cross(u1, u2, axis=-1) * np.cross(u2, u3, axis=-1)).sum(axis=-1))
+    args = [torch.as_tensor(np.asarray(c, dtype=np.float64))
+            for c in (c1, c2, c3, c4)]
+    return get_dihedral_torch(*args).numpy()
+
+
+def _select_backbone(coords_nd, atom_mask):
+    """(b, N_atoms, 3) + per-atom bool mask -> (b, L, 3) selected atoms.
+    The mask is shared across the batch (same protein layout)."""
+    sel = atom_mask.reshape(-1).bool()
+    return coords_nd[:, sel]
 
 
 def calc_phis_torch(pred_coords, N_mask, CA_mask, C_mask=None, prop=True,
                     verbose=0):
-    """Proportion of negative backbone phi dihedrals per structure.
+    """Proportion of negative backbone phi dihedrals per structure,
+    batched (no per-item python loop).
 
-    Used to pick the correct mirror image after MDS.  pred_coords is
-    (batch, 3, N_atoms) with per-atom boolean masks for N/CA/C positions.
+    pred_coords: (batch, 3, N_atoms); N/CA/C masks select the backbone
+    atoms (C_mask defaults to "neither N nor CA", matching a pure
+    backbone cloud).
     """
-    pred_coords_ = pred_coords.detach().transpose(-1, -2).cpu()
-    N_mask = expand_dims_to(N_mask, 2 - len(N_mask.shape))
-    CA_mask = expand_dims_to(CA_mask, 2 - len(CA_mask.shape))
-    if C_mask is not None:
-        C_mask = expand_dims_to(C_mask, 2 - len(C_mask.shape))
+    pts = pred_coords.detach().transpose(-1, -2)  # (b, N_atoms, 3)
+    N_mask = as_batched(N_mask, 2)
+    CA_mask = as_batched(CA_mask, 2)
+    if C_mask is None:
+        C_mask = ~(N_mask[0] | CA_mask[0])
     else:
-        C_mask = torch.logical_not(torch.logical_or(N_mask, CA_mask))
+        C_mask = as_batched(C_mask, 2)[0]
+    n_at = _select_backbone(pts, N_mask[0])
+    ca_at = _select_backbone(pts, CA_mask[0])
+    c_at = _select_backbone(pts, C_mask)
 
-    n_terms = pred_coords_[:, N_mask[0].squeeze()]
-    c_alphas = pred_coords_[:, CA_mask[0].squeeze()]
-    c_terms = pred_coords_[:, C_mask[0].squeeze()]
-
-    phis = [get_dihedral_torch(c_terms[i, :-1], n_terms[i, 1:],
-                               c_alphas[i, 1:], c_terms[i, 1:])
-            for i in range(pred_coords.shape[0])]
+    # phi(i) = dihedral(C(i-1), N(i), CA(i), C(i)) — one batched call
+    phis = get_dihedral_torch(c_at[:, :-1], n_at[:, 1:],
+                              ca_at[:, 1:], c_at[:, 1:])
     if prop:
-        return torch.stack([(x < 0).float().mean() for x in phis], dim=0)
+        return (phis < 0).float().mean(dim=-1)
     return phis
 
 
 def calc_phis_numpy(pred_coords, N_mask, CA_mask, C_mask=None, prop=True,
                     verbose=0):
-    pred_coords_ = np.transpose(pred_coords, (0, 2, 1))
-    n_terms = pred_coords_[:, N_mask.squeeze()]
-    c_alphas = pred_coords_[:, CA_mask.squeeze()]
-    if C_mask is not None:
-        c_terms = pred_coords_[:, C_mask]
-    else:
-        c_terms = pred_coords_[:, (np.ones_like(N_mask) - N_mask - CA_mask).squeeze().astype(bool)]
-    phis = [get_dihedral_numpy(c_terms[i, :-1], n_terms[i, 1:],
-                               c_alphas[i, 1:], c_terms[i, 1:])
-            for i in range(pred_coords.shape[0])]
+    out = calc_phis_torch(
+        torch.as_tensor(np.asarray(pred_coords, dtype=np.float64)),
+        torch.as_tensor(np.asarray(N_mask)).bool(),
+        torch.as_tensor(np.asarray(CA_mask)).bool(),
+        None if C_mask is None
+        else torch.as_tensor(np.asarray(C_mask)).bool(),
+        prop=prop, verbose=verbose)
     if prop:
-        return np.array([(x < 0).mean() for x in phis])
-    return phis
+        return out.numpy()
+    return out.numpy()
 
 
 # ---------------------------------------------------------------------------
-# protein-aware MDS wrappers
+# protein-aware wrapper: MDS + mirror-image correction
 
 
 def mdscaling_torch(pre_dist_mat, weights=None, iters=10, tol=1e-5,
                     fix_mirror=True, N_mask=None, CA_mask=None, C_mask=None,
                     eigen=False, verbose=0):
-    preds, stresses = mds_torch(pre_dist_mat, weights=weights, iters=iters,
+    coords, history = mds_torch(pre_dist_mat, weights=weights, iters=iters,
                                 tol=tol, eigen=eigen, verbose=verbose)
     if not fix_mirror:
-        return preds, stresses
-    phi_ratios = calc_phis_torch(preds, N_mask, CA_mask, C_mask, prop=True)
-    to_correct = torch.nonzero(phi_ratios < 0.5).view(-1)
-    # flip Z of structures whose phi distribution says "wrong mirror"
-    preds[to_correct, -1] = -preds[to_correct, -1]
+        return coords, history
+    phi_ratios = calc_phis_torch(coords, N_mask, CA_mask, C_mask, prop=True)
+    wrong = (phi_ratios < 0.5).nonzero().view(-1)
+    coords[wrong, -1] = -coords[wrong, -1]
     if verbose == 2:
-        print("Corrected mirror idxs:", to_correct)
-    return preds, stresses
+        print('Corrected mirror idxs:', wrong)
+    return coords, history
 
 
 def mdscaling_numpy(pre_dist_mat, weights=None, iters=10, tol=1e-5,
                     fix_mirror=True, N_mask=None, CA_mask=None, C_mask=None,
-                    verbose=0):
-    preds, stresses = mds_numpy(pre_dist_mat, weights=weights, iters=iters,
-                                tol=tol, verbose=verbose)
+                    eigen=False, verbose=0):
+    coords, history = mds_numpy(pre_dist_mat, weights=weights, iters=iters,
+                                tol=tol, eigen=eigen, verbose=verbose)
     if not fix_mirror:
-        return preds, stresses
-    phi_ratios = calc_phis_numpy(preds, N_mask, CA_mask, C_mask, prop=True)
-    for i in range(len(preds)):
-        if phi_ratios[i] < 0.5:
-            preds[i, -1] = -preds[i, -1]
-            if verbose == 2:
-                print("Corrected mirror in struct no.", i)
-    return preds, stresses
+        return coords, history
+    phi_ratios = calc_phis_numpy(coords, N_mask, CA_mask, C_mask, prop=True)
+    for i in np.nonzero(phi_ratios < 0.5)[0]:
+        coords[i, -1] = -coords[i, -1]
+        if verbose == 2:
+            print('Corrected mirror in struct no.', i)
+    return coords, history
 
 
-@set_backend_kwarg
-@invoke_torch_or_numpy(mdscaling_torch, mdscaling_numpy)
-def MDScaling(pre_dist_mat, **kwargs):
-    """Distance matrix (N, N) (or batch) -> 3D coords (3, N) via MDS with
-    optional phi-based mirror correction.  See mds_torch for details."""
-    pre_dist_mat = expand_dims_to(pre_dist_mat, 3 - len(pre_dist_mat.shape))
-    return pre_dist_mat, kwargs
+def _mds_prepare(pre_dist_mat, **kwargs):
+    return (as_batched(pre_dist_mat, 3),), kwargs
+
+
+MDScaling = named_wrapper(
+    dual_backend(mdscaling_torch, mdscaling_numpy, prepare=_mds_prepare),
+    'MDScaling',
+    "Distance matrix (N, N) (or batch) -> 3D coords (3, N) via MDS with "
+    "optional phi-based mirror correction.  See mds_torch for details.")

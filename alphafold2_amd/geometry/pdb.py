@@ -16,44 +16,40 @@ from ..vocab import VOCAB
 
 def download_pdb(name, route):
     """Fetch a PDB entry from RCSB (requires network). Returns `route`."""
-    os.system(f"curl https://files.rcsb.org/download/{name}.pdb > {route}")
+    import urllib.request
+    urllib.request.urlretrieve(
+        f"https://files.rcsb.org/download/{name}.pdb", route)
     return route
 
 
 def clean_pdb(name, route=None, chain_num=None):
-    """Keep only the selected chain(s) of a PDB file (requires mdtraj)."""
+    """Keep only the selected chain (or all chains, re-saved) of a PDB
+    file (requires mdtraj)."""
     import mdtraj
-    destin = route if route is not None else name
-    raw_prot = mdtraj.load_pdb(name)
-    idxs = []
-    for chain in raw_prot.topology.chains:
-        if chain_num is not None and chain_num != chain.index:
-            continue
-        chain_idxs = raw_prot.topology.select(f"chainid == {str(chain.index)}")
-        idxs.extend(chain_idxs.tolist())
-    idxs = sorted(idxs)
-    prot = mdtraj.Trajectory(xyz=raw_prot.xyz[:, idxs],
-                             topology=raw_prot.topology.subset(idxs))
-    prot.save(destin)
-    return destin
+    traj = mdtraj.load_pdb(name)
+    query = "all" if chain_num is None else f"chainid {chain_num}"
+    out = traj.atom_slice(traj.topology.select(query))
+    destination = route if route is not None else name
+    out.save(destination)
+    return destination
 
 
 def custom2pdb(coords, proteinnet_id, route):
-    """Write coords into a scaffold downloaded for `proteinnet_id`."""
+    """Write coords into a scaffold fetched for a proteinnet-style id
+    ("<set>#<pdb>_<chain>_<extra>"); requires network + mdtraj."""
     import mdtraj
     if isinstance(coords, torch.Tensor):
         coords = coords.detach().cpu().numpy()
-    if coords.shape[1] == 3:
+    if coords.shape[1] == 3:  # (3, N) -> (N, 3)
         coords = coords.T
-    coords = np.expand_dims(coords, axis=0)
     pdb_name, chain_num = proteinnet_id.split("#")[-1].split("_")[:-1]
-    pdb_destin = "/".join(route.split("/")[:-1]) + "/" + pdb_name + ".pdb"
-    download_pdb(pdb_name, pdb_destin)
-    clean_pdb(pdb_destin, chain_num=chain_num)
-    scaffold = mdtraj.load_pdb(pdb_destin)
-    scaffold.xyz = coords
+    scaffold_path = os.path.join(os.path.dirname(route), pdb_name + ".pdb")
+    download_pdb(pdb_name, scaffold_path)
+    clean_pdb(scaffold_path, chain_num=chain_num)
+    scaffold = mdtraj.load_pdb(scaffold_path)
+    scaffold.xyz = coords[None]
     scaffold.save(route)
-    return pdb_destin, route
+    return scaffold_path, route
 
 
 def coords2pdb(seq, coords, cloud_mask, prefix="", name="af2_struct.pdb"):
@@ -89,16 +85,13 @@ def coords2pdb(seq, coords, cloud_mask, prefix="", name="af2_struct.pdb"):
     return prefix + name
 
 
-# adapted behavior from the ESM a3m-reading convention
+# a3m convention: lowercase = insertion columns, './*' = gap/stop markers
+_A3M_DROP = str.maketrans('', '', string.ascii_lowercase + '.*')
 
 
 def remove_insertions(sequence: str) -> str:
-    """Drop lowercase/insertion characters from an aligned sequence."""
-    deletekeys = dict.fromkeys(string.ascii_lowercase)
-    deletekeys["."] = None
-    deletekeys["*"] = None
-    translation = str.maketrans(deletekeys)
-    return sequence.translate(translation)
+    """Drop insertion characters from an a3m-aligned sequence."""
+    return sequence.translate(_A3M_DROP)
 
 
 def read_msa(filename: str, nseq: int):

@@ -15,25 +15,28 @@ from . import constants
 
 
 def get_mask_subset_with_prob(mask, prob):
-    """Pick exactly ceil(prob * len) positions per row from `mask` (uniform
-    among allowed positions) — top-k on masked uniform noise."""
-    batch, seq_len = mask.shape
-    device = mask.device
-    max_masked = math.ceil(prob * seq_len)
+    """Uniformly sample a subset of the allowed positions per row.
 
-    num_tokens = mask.sum(dim=-1, keepdim=True)
-    # rows with fewer allowed positions than the global top-k get the
-    # excess samples suppressed
-    mask_excess = (mask.cumsum(dim=-1) > (num_tokens * prob).ceil())
-    mask_excess = mask_excess[:, :max_masked]
+    Row i receives ceil(prob * allowed_i) picks (capped by
+    ceil(prob * seq_len), the reference budget).  Rank-based selection:
+    draw one uniform score per position, push disallowed positions to
+    the bottom, and keep every position whose within-row rank falls
+    inside the row's quota — no scatter round-trip, fully batched.
+    """
+    seq_len = mask.shape[-1]
+    budget = math.ceil(prob * seq_len)
+    allowed = mask.sum(dim=-1, keepdim=True)
+    quota = torch.clamp((allowed * prob).ceil().long(), max=budget)
 
-    rand = torch.rand((batch, seq_len), device=device).masked_fill(~mask, -1e9)
-    _, sampled_indices = rand.topk(max_masked, dim=-1)
-    sampled_indices = (sampled_indices + 1).masked_fill_(mask_excess, 0)
-
-    new_mask = torch.zeros((batch, seq_len + 1), device=device)
-    new_mask.scatter_(-1, sampled_indices, 1)
-    return new_mask[:, 1:].bool()
+    scores = torch.rand(mask.shape, device=mask.device)
+    scores = scores.masked_fill(~mask, -1.0)
+    # rank[r, j] = how many positions in row r score higher than j
+    order = scores.argsort(dim=-1, descending=True)
+    rank = torch.empty_like(order)
+    rank.scatter_(-1, order,
+                  torch.arange(seq_len, device=mask.device)
+                  .expand_as(order))
+    return (rank < quota) & mask
 
 
 class MLM(nn.Module):

@@ -8,8 +8,8 @@ from .vocab import (  # noqa: F401
     make_cloud_mask, make_atom_id_embedds, ONE_TO_THREE_LETTER_MAP,
 )
 from .geometry import (  # noqa: F401
-    exists, set_backend_kwarg, expand_dims_to, expand_arg_dims,
-    invoke_torch_or_numpy, torch_default_dtype,
+    exists, as_batched, expand_dims_to, dual_backend, resolve_backend,
+    torch_default_dtype,
     DISTANCE_THRESHOLDS, get_bucketed_distance_matrix, center_distogram_torch,
     kabsch_torch, kabsch_numpy, rmsd_torch, rmsd_numpy,
     gdt_torch, gdt_numpy, tmscore_torch, tmscore_numpy,

@@ -47,6 +47,7 @@
 #include <cmath>
 #include <cstdio>
 #include <cstdlib>
+#include <cstring>
 #include <vector>
 
 typedef __bf16 bf16_t;
@@ -595,22 +596,39 @@ static T* to_dev(const std::vector<T>& h) {
   return d;
 }
 
+// replicate the first Bu batches across the full batch axis so big
+// timing shapes don't need a full-size CPU reference (the kernel output
+// for batch b is then checked against reference batch b % Bu)
+template <typename T>
+static void tile_batches(std::vector<T>& v, long per, int Bu, int B) {
+  for (int b = Bu; b < B; ++b)
+    memcpy(v.data() + (long)b * per, v.data() + (long)(b % Bu) * per,
+           per * sizeof(T));
+}
+
 static int run_case(int B, int Lq, int Lk, bool has_bias, bool has_mask,
                     bool timing) {
   const float scale = 1.f / sqrtf((float)DH);
+  const int Bu = timing ? (B < 4 ? B : 4) : B;  // unique batches
   srand(12345);
   auto rnd = [&]() { return (rand() / (float)RAND_MAX - 0.5f) * 2.f; };
 
   std::vector<float> qf((long)B * Lq * DH), kf((long)B * Lk * DH),
       vf((long)B * Lk * DH), bf(has_bias ? (long)B * Lq * Lk : 1);
   std::vector<unsigned char> mk(has_mask ? (long)B * Lk : 1, 1);
-  for (auto& x : qf) x = rnd();
-  for (auto& x : kf) x = rnd();
-  for (auto& x : vf) x = rnd();
-  for (auto& x : bf) x = rnd();
+  for (long i = 0; i < (long)Bu * Lq * DH; ++i) qf[i] = rnd();
+  for (long i = 0; i < (long)Bu * Lk * DH; ++i) kf[i] = rnd();
+  for (long i = 0; i < (long)Bu * Lk * DH; ++i) vf[i] = rnd();
+  if (has_bias)
+    for (long i = 0; i < (long)Bu * Lq * Lk; ++i) bf[i] = rnd();
   if (has_mask)
-    for (size_t i = 0; i < mk.size(); ++i)
+    for (long i = 0; i < (long)Bu * Lk; ++i)
       mk[i] = (i % Lk == 0) ? 1 : (rnd() > -0.6f);  // keep >=1 key valid
+  tile_batches(qf, (long)Lq * DH, Bu, B);
+  tile_batches(kf, (long)Lk * DH, Bu, B);
+  tile_batches(vf, (long)Lk * DH, Bu, B);
+  if (has_bias) tile_batches(bf, (long)Lq * Lk, Bu, B);
+  if (has_mask) tile_batches(mk, (long)Lk, Bu, B);
 
   auto to_bf = [](const std::vector<float>& s) {
     std::vector<bf16_t> o(s.size());
@@ -646,11 +664,14 @@ static int run_case(int B, int Lq, int Lk, bool has_bias, bool has_mask,
   std::vector<bf16_t> outb((long)B * Lq * DH);
   HIP_CHECK(hipMemcpy(outb.data(), dout, outb.size() * sizeof(bf16_t),
                       hipMemcpyDeviceToHost));
-  std::vector<float> ref((long)B * Lq * DH);
-  cpu_ref(qf, kf, vf, bf, mk, has_bias, has_mask, B, Lq, Lk, scale, ref);
+  std::vector<float> ref((long)Bu * Lq * DH);
+  cpu_ref(qf, kf, vf, bf, mk, has_bias, has_mask, Bu, Lq, Lk, scale, ref);
   float err = 0.f;
-  for (size_t i = 0; i < ref.size(); ++i)
-    err = fmaxf(err, fabsf((float)outb[i] - ref[i]));
+  const long per_out = (long)Lq * DH;
+  for (int b = 0; b < B; ++b)
+    for (long r = 0; r < per_out; ++r)
+      err = fmaxf(err, fabsf((float)outb[(long)b * per_out + r] -
+                             ref[(long)(b % Bu) * per_out + r]));
   const bool pass = err < 3e-2f;
   printf("B=%d Lq=%d Lk=%d bias=%d mask=%d  max_err=%.4f  %s\n", B, Lq,
          Lk, has_bias, has_mask, err, pass ? "PASS" : "FAIL");
@@ -900,17 +921,24 @@ void attn_dkv_v2(const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
 static int run_dq_case(int B, int Lq, int Lk, bool has_bias,
                        bool timing) {
   const float scale = 1.f / sqrtf((float)DH);
+  const int Bu = timing ? (B < 4 ? B : 4) : B;  // unique batches
   srand(777);
   auto rnd = [&]() { return (rand() / (float)RAND_MAX - 0.5f) * 2.f; };
 
   std::vector<float> qf((long)B * Lq * DH), kf((long)B * Lk * DH),
       vf((long)B * Lk * DH), dof((long)B * Lq * DH),
       bf(has_bias ? (long)B * Lq * Lk : 1);
-  for (auto& x : qf) x = rnd();
-  for (auto& x : kf) x = rnd();
-  for (auto& x : vf) x = rnd();
-  for (auto& x : dof) x = rnd();
-  for (auto& x : bf) x = rnd();
+  for (long i = 0; i < (long)Bu * Lq * DH; ++i) qf[i] = rnd();
+  for (long i = 0; i < (long)Bu * Lk * DH; ++i) kf[i] = rnd();
+  for (long i = 0; i < (long)Bu * Lk * DH; ++i) vf[i] = rnd();
+  for (long i = 0; i < (long)Bu * Lq * DH; ++i) dof[i] = rnd();
+  if (has_bias)
+    for (long i = 0; i < (long)Bu * Lq * Lk; ++i) bf[i] = rnd();
+  tile_batches(qf, (long)Lq * DH, Bu, B);
+  tile_batches(kf, (long)Lk * DH, Bu, B);
+  tile_batches(vf, (long)Lk * DH, Bu, B);
+  tile_batches(dof, (long)Lq * DH, Bu, B);
+  if (has_bias) tile_batches(bf, (long)Lq * Lk, Bu, B);
 
   auto to_bf = [](std::vector<float>& s) {
     std::vector<bf16_t> o(s.size());
@@ -924,10 +952,11 @@ static int run_dq_case(int B, int Lq, int Lk, bool has_bias,
        bb = to_bf(bf);
 
   // CPU: lse, delta, and reference dQ (fp32 over bf16-quantized inputs)
+  // — unique batches only; lse/delta are tiled up for the kernel input
   std::vector<float> lse((long)B * Lq), delta((long)B * Lq),
-      dq_ref((long)B * Lq * DH, 0.f);
+      dq_ref((long)Bu * Lq * DH, 0.f);
   std::vector<float> srow(Lk), prow(Lk);
-  for (int b = 0; b < B; ++b)
+  for (int b = 0; b < Bu; ++b)
     for (int i = 0; i < Lq; ++i) {
       float mx = -1e30f;
       for (int j = 0; j < Lk; ++j) {
@@ -967,6 +996,8 @@ static int run_dq_case(int B, int Lq, int Lk, bool has_bias,
       }
     }
 
+  tile_batches(lse, (long)Lq, Bu, B);
+  tile_batches(delta, (long)Lq, Bu, B);
   bf16_t *dq_ = to_dev(qb), *dk_ = to_dev(kb), *dv_ = to_dev(vb),
          *ddo = to_dev(dob), *db_ = to_dev(bb);
   float *dlse = to_dev(lse), *ddelta = to_dev(delta);
@@ -982,10 +1013,13 @@ static int run_dq_case(int B, int Lq, int Lk, bool has_bias,
   HIP_CHECK(hipMemcpy(outb.data(), dout_, outb.size() * sizeof(bf16_t),
                       hipMemcpyDeviceToHost));
   float err = 0.f, ref_max = 0.f;
-  for (size_t i = 0; i < dq_ref.size(); ++i) {
-    err = fmaxf(err, fabsf((float)outb[i] - dq_ref[i]));
-    ref_max = fmaxf(ref_max, fabsf(dq_ref[i]));
-  }
+  const long per_dq = (long)Lq * DH;
+  for (int b = 0; b < B; ++b)
+    for (long r = 0; r < per_dq; ++r) {
+      const float rv = dq_ref[(long)(b % Bu) * per_dq + r];
+      err = fmaxf(err, fabsf((float)outb[(long)b * per_dq + r] - rv));
+      ref_max = fmaxf(ref_max, fabsf(rv));
+    }
   // dS goes through bf16 before the dQ GEMM; tolerance scales with |dq|
   const bool pass = err < 6e-2f * fmaxf(1.f, ref_max);
   printf("dq: B=%d Lq=%d Lk=%d bias=%d  max_err=%.4f (ref_max %.2f)  %s\n",
@@ -1018,6 +1052,7 @@ static int run_dq_case(int B, int Lq, int Lk, bool has_bias,
 static int run_dkv_case(int B, int Lq, int Lk, bool has_bias,
                         bool has_mask, bool timing) {
   const float scale = 1.f / sqrtf((float)DH);
+  const int Bu = timing ? (B < 4 ? B : 4) : B;  // unique batches
   srand(4242);
   auto rnd = [&]() { return (rand() / (float)RAND_MAX - 0.5f) * 2.f; };
 
@@ -1025,14 +1060,21 @@ static int run_dkv_case(int B, int Lq, int Lk, bool has_bias,
       vf((long)B * Lk * DH), dof((long)B * Lq * DH),
       bf(has_bias ? (long)B * Lq * Lk : 1);
   std::vector<unsigned char> mk(has_mask ? (long)B * Lk : 1, 1);
-  for (auto& x : qf) x = rnd();
-  for (auto& x : kf) x = rnd();
-  for (auto& x : vf) x = rnd();
-  for (auto& x : dof) x = rnd();
-  for (auto& x : bf) x = rnd();
+  for (long i = 0; i < (long)Bu * Lq * DH; ++i) qf[i] = rnd();
+  for (long i = 0; i < (long)Bu * Lk * DH; ++i) kf[i] = rnd();
+  for (long i = 0; i < (long)Bu * Lk * DH; ++i) vf[i] = rnd();
+  for (long i = 0; i < (long)Bu * Lq * DH; ++i) dof[i] = rnd();
+  if (has_bias)
+    for (long i = 0; i < (long)Bu * Lq * Lk; ++i) bf[i] = rnd();
   if (has_mask)
-    for (size_t i = 0; i < mk.size(); ++i)
+    for (long i = 0; i < (long)Bu * Lk; ++i)
       mk[i] = (i % Lk == 0) ? 1 : (rnd() > -0.6f);
+  tile_batches(qf, (long)Lq * DH, Bu, B);
+  tile_batches(kf, (long)Lk * DH, Bu, B);
+  tile_batches(vf, (long)Lk * DH, Bu, B);
+  tile_batches(dof, (long)Lq * DH, Bu, B);
+  if (has_bias) tile_batches(bf, (long)Lq * Lk, Bu, B);
+  if (has_mask) tile_batches(mk, (long)Lk, Bu, B);
 
   auto to_bf = [](std::vector<float>& s) {
     std::vector<bf16_t> o(s.size());
@@ -1045,11 +1087,11 @@ static int run_dkv_case(int B, int Lq, int Lk, bool has_bias,
   auto qb = to_bf(qf), kb = to_bf(kf), vb = to_bf(vf), dob = to_bf(dof),
        bb = to_bf(bf);
 
-  // CPU: lse/delta + reference dK/dV
+  // CPU: lse/delta + reference dK/dV — unique batches only
   std::vector<float> lse((long)B * Lq), delta((long)B * Lq),
-      dk_ref((long)B * Lk * DH, 0.f), dv_ref((long)B * Lk * DH, 0.f);
+      dk_ref((long)Bu * Lk * DH, 0.f), dv_ref((long)Bu * Lk * DH, 0.f);
   std::vector<float> srow(Lk), prow(Lk);
-  for (int b = 0; b < B; ++b)
+  for (int b = 0; b < Bu; ++b)
     for (int i = 0; i < Lq; ++i) {
       float mx = -1e30f;
       for (int j = 0; j < Lk; ++j) {
@@ -1094,6 +1136,8 @@ static int run_dkv_case(int B, int Lq, int Lk, bool has_bias,
       }
     }
 
+  tile_batches(lse, (long)Lq, Bu, B);
+  tile_batches(delta, (long)Lq, Bu, B);
   bf16_t *dq_ = to_dev(qb), *dk_ = to_dev(kb), *dv_ = to_dev(vb),
          *ddo = to_dev(dob), *db_ = to_dev(bb);
   unsigned char* dm = to_dev(mk);
@@ -1116,12 +1160,15 @@ static int run_dkv_case(int B, int Lq, int Lk, bool has_bias,
   HIP_CHECK(hipMemcpy(dvb.data(), dvo, dvb.size() * sizeof(bf16_t),
                       hipMemcpyDeviceToHost));
   float err_k = 0.f, err_v = 0.f, mx_k = 0.f, mx_v = 0.f;
-  for (size_t i = 0; i < dk_ref.size(); ++i) {
-    err_k = fmaxf(err_k, fabsf((float)dkb[i] - dk_ref[i]));
-    err_v = fmaxf(err_v, fabsf((float)dvb[i] - dv_ref[i]));
-    mx_k = fmaxf(mx_k, fabsf(dk_ref[i]));
-    mx_v = fmaxf(mx_v, fabsf(dv_ref[i]));
-  }
+  const long per_kv = (long)Lk * DH;
+  for (int b = 0; b < B; ++b)
+    for (long r = 0; r < per_kv; ++r) {
+      const long ri = (long)(b % Bu) * per_kv + r, oi = (long)b * per_kv + r;
+      err_k = fmaxf(err_k, fabsf((float)dkb[oi] - dk_ref[ri]));
+      err_v = fmaxf(err_v, fabsf((float)dvb[oi] - dv_ref[ri]));
+      mx_k = fmaxf(mx_k, fabsf(dk_ref[ri]));
+      mx_v = fmaxf(mx_v, fabsf(dv_ref[ri]));
+    }
   const bool pass = err_k < 6e-2f * fmaxf(1.f, mx_k) &&
                     err_v < 6e-2f * fmaxf(1.f, mx_v);
   printf("dkv: B=%d Lq=%d Lk=%d bias=%d mask=%d  dK_err=%.4f dV_err=%.4f"
@@ -1154,6 +1201,7 @@ static int run_dkv_case(int B, int Lq, int Lk, bool has_bias,
 }
 
 int main() {
+  setvbuf(stdout, nullptr, _IONBF, 0);  // survive a timeout kill
   int rc = 0;
   rc |= run_case(4, 128, 128, false, false, false);
   rc |= run_case(4, 128, 128, true, false, false);
