@@ -71,9 +71,14 @@ class FeedForward(nn.Module):
         )
         init_zero_(self.net[-1])
 
-    def forward(self, x, **kwargs):
+    def forward(self, x, residual=None, **kwargs):
+        """`residual`, when given, is added inside the output GEMM's
+        epilogue (saves a full elementwise pass over the tensor)."""
         x = self.norm(x)
-        return self.net(x)
+        h = ops.ff1_geglu(x, self.net[0].weight, self.net[0].bias)
+        h = self.net[2](h)
+        return ops.fused_linear(h, self.net[3].weight, self.net[3].bias,
+                                residual=residual)
 
 
 # ---------------------------------------------------------------------------
@@ -389,18 +394,20 @@ class EvoformerBlock(nn.Module):
             and torch.is_grad_enabled()
 
         def run_ff(f, t):
+            # residual folded into the FF output GEMM's epilogue
             if ckpt_ff:
-                return checkpoint(f, t, use_reentrant=False,
+                return checkpoint(lambda u: f(u, residual=u), t,
+                                  use_reentrant=False,
                                   preserve_rng_state=self.ff_dropout_p > 0)
-            return f(t)
+            return f(t, residual=t)
 
         # MSA attention and transition
         m = msa_attn(m, mask=msa_mask, pairwise_repr=x)
-        m = run_ff(msa_ff, m) + m
+        m = run_ff(msa_ff, m)
 
         # pairwise attention and transition
         x = attn(x, mask=mask, msa_repr=m, msa_mask=msa_mask)
-        x = run_ff(ff, x) + x
+        x = run_ff(ff, x)
 
         return x, m, mask, msa_mask
 

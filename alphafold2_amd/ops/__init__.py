@@ -24,4 +24,5 @@ from .dispatch import (  # noqa: F401
     hip_ops_available, using_hip,
     attention_core, attention_core_packed, geglu, outer_product_mean, triangle_mix,
     pair_outer_sum, distance_buckets, layer_norm, softclamp_gate,
+    fused_linear, ff1_geglu,
 )

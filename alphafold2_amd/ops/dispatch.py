@@ -117,6 +117,34 @@ def geglu(x):
     return eager.geglu(x)
 
 
+def fused_linear(x, weight, bias=None, residual=None):
+    """out = x @ weight.T (+ bias) (+ residual) through the custom
+    tall-M/small-K MFMA GEMM (ffgemm.hip) when fusable."""
+    ok = (x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
+          and x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
+          and (residual is None or residual.dtype == torch.bfloat16)
+          and using_hip(x, 'linear_fwd'))
+    if ok:
+        from .hip_autograd import hip_linear
+        return hip_linear(x, weight, bias, residual)
+    out = torch.nn.functional.linear(x, weight, bias)
+    if residual is not None:
+        out = out + residual
+    return out
+
+
+def ff1_geglu(x, weight, bias=None):
+    """GEGLU(x @ weight.T + bias) with the GEMM + chunk + gelu + mul
+    fused into one kernel (K6 of SURVEY.md §2.17)."""
+    ok = (x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
+          and x.shape[-1] % 8 == 0 and weight.shape[0] % 16 == 0
+          and using_hip(x, 'ff1_geglu_fwd'))
+    if ok:
+        from .hip_autograd import hip_ff1_geglu
+        return hip_ff1_geglu(x, weight, bias)
+    return geglu(torch.nn.functional.linear(x, weight, bias))
+
+
 def outer_product_mean(left, right, mask=None, eps=1e-5):
     if left.dtype == torch.bfloat16 and using_hip(left, 'pcgemm'):
         from .hip_autograd import hip_outer_product_mean

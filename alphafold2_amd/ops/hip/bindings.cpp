@@ -20,6 +20,11 @@ at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs,
 std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x, at::Tensor g,
                                     long xs, long gs,
                                     c10::optional<at::Tensor> rowmask);
+at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
+                      c10::optional<at::Tensor> bias,
+                      c10::optional<at::Tensor> resid);
+std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
+                                      c10::optional<at::Tensor> bias);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
@@ -48,6 +53,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gatemul_bwd", &gatemul_bwd, "fused x*sigmoid(g) backward (gfx950)",
         py::arg("dy"), py::arg("x"), py::arg("g"), py::arg("xs"),
         py::arg("gs"), py::arg("rowmask") = c10::nullopt);
+  m.def("linear_fwd", &linear_fwd,
+        "tall-M small-K linear GEMM, bias/residual epilogue (gfx950 MFMA)",
+        py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt,
+        py::arg("resid") = c10::nullopt);
+  m.def("ff1_geglu_fwd", &ff1_geglu_fwd,
+        "linear GEMM with fused GEGLU epilogue (gfx950 MFMA)",
+        py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt);
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (gfx950)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
         py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"));

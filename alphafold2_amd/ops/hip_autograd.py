@@ -44,6 +44,91 @@ def hip_geglu(x):
     return _GegluFn.apply(x)
 
 
+def _dgrad(dy2, weight):
+    """dX = dY @ W through the custom GEMM (A row-major, weight given
+    as (N, K): dgrad needs (K, N) = W.T, a small contiguous copy)."""
+    ext = _load_ext()
+    wt = weight.t().contiguous()
+    if dy2.shape[-1] % 8 == 0 and wt.shape[0] % 8 == 0:
+        return ext.linear_fwd(dy2, wt, None, None)
+    return dy2 @ weight
+
+
+class _LinearFn(torch.autograd.Function):
+    """out = x @ W.T (+ bias) (+ residual) — forward on the fused MFMA
+    GEMM; dgrad reuses the same kernel, wgrad stays on hipBLASLt (its
+    K = M reduction is a different shape class)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, residual):
+        ext = _load_ext()
+        x = x.contiguous()
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        ctx.has_resid = residual is not None
+        return ext.linear_fwd(
+            x, weight.contiguous(),
+            bias.contiguous() if bias is not None else None,
+            residual.contiguous() if residual is not None else None)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy2 = dy.contiguous().reshape(-1, dy.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = _dgrad(dy2, weight).view_as(x)
+        if ctx.needs_input_grad[1]:
+            dw = dy2.t() @ x2
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy2.sum(dim=0)
+        dr = dy if ctx.has_resid else None
+        return dx, dw, db, dr
+
+
+def hip_linear(x, weight, bias=None, residual=None):
+    return _LinearFn.apply(x, weight, bias, residual)
+
+
+class _FF1GegluFn(torch.autograd.Function):
+    """GEGLU(x @ W.T + bias): one fused GEMM writing the gated half-width
+    output plus the raw pre-activation (backward needs it for the gelu
+    grads — the same tensor today's unfused path stores as the Linear
+    output)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ext = _load_ext()
+        x = x.contiguous()
+        out, inter = ext.ff1_geglu_fwd(
+            x, weight.contiguous(),
+            bias.contiguous() if bias is not None else None)
+        ctx.save_for_backward(x, weight, inter)
+        ctx.has_bias = bias is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _load_ext()
+        x, weight, inter = ctx.saved_tensors
+        di = ext.geglu_bwd(dy.contiguous(), inter)  # (..., N) pre-act grad
+        di2 = di.reshape(-1, di.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = _dgrad(di2, weight).view_as(x)
+        if ctx.needs_input_grad[1]:
+            dw = di2.t() @ x2
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = di2.sum(dim=0)
+        return dx, dw, db
+
+
+def hip_ff1_geglu(x, weight, bias=None):
+    return _FF1GegluFn.apply(x, weight, bias)
+
+
 class _AttentionFn(torch.autograd.Function):
     """Fused flash attention (bf16, head dim 64) with broadcast pair bias.
 
