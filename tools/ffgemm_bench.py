@@ -1,0 +1,74 @@
+"""Within-run A/B of the fused linear GEMM (ffgemm.hip) vs hipBLASLt on
+the production FF/projection shapes (flagship config dim=256, crop 256,
+msa 128, batch 5).  Run on a GPU box:
+
+    python tools/ffgemm_bench.py
+"""
+import torch
+
+from alphafold2_amd.ops.dispatch import _load_ext
+
+
+def time_fn(fn, iters=30, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = torch.cuda.Event(enable_timing=True)
+    t1 = torch.cuda.Event(enable_timing=True)
+    t0.record()
+    for _ in range(iters):
+        fn()
+    t1.record()
+    t1.synchronize()
+    return t0.elapsed_time(t1) / iters
+
+
+def main():
+    ext = _load_ext()
+    assert ext is not None
+    torch.manual_seed(0)
+    dev = 'cuda'
+    dt = torch.bfloat16
+
+    # (name, M, K, N, epi)
+    shapes = [
+        ('msa_ff1_geglu', 5 * 128 * 256, 256, 4096, 'geglu'),
+        ('pair_ff1_geglu', 5 * 256 * 256, 256, 4096, 'geglu'),
+        ('msa_ff2_resid', 5 * 128 * 256, 1024, 256, 'resid'),
+        ('pair_ff2_resid', 5 * 256 * 256, 1024, 256, 'resid'),
+        ('attn_qkvg', 5 * 256 * 256, 256, 2048, 'none'),
+        ('attn_out', 5 * 256 * 256, 512, 256, 'none'),
+        ('trimul_proj', 5 * 256 * 256, 256, 640, 'none'),
+        ('dgrad_ff1', 5 * 128 * 256, 2048, 256, 'none'),
+    ]
+    for name, M, K, N, epi in shapes:
+        x = torch.randn(M, K, device=dev, dtype=dt)
+        w = torch.randn(N, K, device=dev, dtype=dt) * 0.1
+        b = torch.randn(N, device=dev, dtype=dt)
+        r = torch.randn(M, N, device=dev, dtype=dt) if epi == 'resid' \
+            else None
+        fl = 2.0 * M * K * N
+
+        t_blas = time_fn(lambda: torch.nn.functional.linear(x, w, b))
+        if epi == 'geglu':
+            t_mine = time_fn(lambda: ext.ff1_geglu_fwd(x, w, b))
+            # reference composition cost: linear + geglu
+            t_ref2 = time_fn(lambda: ext.geglu_fwd(
+                torch.nn.functional.linear(x, w, b)))
+            extra = f'  blas+geglu={t_ref2:.3f}ms'
+        elif epi == 'resid':
+            t_mine = time_fn(lambda: ext.linear_fwd(x, w, b, r))
+            t_ref2 = time_fn(
+                lambda: torch.nn.functional.linear(x, w, b) + r)
+            extra = f'  blas+add={t_ref2:.3f}ms'
+        else:
+            t_mine = time_fn(lambda: ext.linear_fwd(x, w, b, None))
+            extra = ''
+        print(f'{name:16s} M={M:7d} K={K:4d} N={N:4d}  '
+              f'blas={t_blas:.3f}ms ({fl / t_blas / 1e9:.0f} TF)  '
+              f'mine={t_mine:.3f}ms ({fl / t_mine / 1e9:.0f} TF)'
+              f'{extra}', flush=True)
+
+
+if __name__ == '__main__':
+    main()
