@@ -117,12 +117,27 @@ def geglu(x):
     return eager.geglu(x)
 
 
+def _bf16_ok(*tensors):
+    """True when every tensor is bf16 already, or will be autocast to
+    bf16 inside the custom_fwd wrapper (fp32 under cuda bf16 autocast)."""
+    amp = (torch.is_autocast_enabled('cuda')
+           and torch.get_autocast_dtype('cuda') == torch.bfloat16)
+    for t in tensors:
+        if t is None:
+            continue
+        if t.dtype == torch.bfloat16:
+            continue
+        if amp and t.dtype in (torch.float32, torch.float16):
+            continue
+        return False
+    return True
+
+
 def fused_linear(x, weight, bias=None, residual=None):
     """out = x @ weight.T (+ bias) (+ residual) through the custom
     tall-M/small-K MFMA GEMM (ffgemm.hip) when fusable."""
-    ok = (x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
-          and x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
-          and (residual is None or residual.dtype == torch.bfloat16)
+    ok = (x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
+          and _bf16_ok(x, weight, residual)
           and using_hip(x, 'linear_fwd'))
     if ok:
         from .hip_autograd import hip_linear
@@ -136,8 +151,8 @@ def fused_linear(x, weight, bias=None, residual=None):
 def ff1_geglu(x, weight, bias=None):
     """GEGLU(x @ weight.T + bias) with the GEMM + chunk + gelu + mul
     fused into one kernel (K6 of SURVEY.md §2.17)."""
-    ok = (x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
-          and x.shape[-1] % 8 == 0 and weight.shape[0] % 16 == 0
+    ok = (x.shape[-1] % 8 == 0 and weight.shape[0] % 16 == 0
+          and _bf16_ok(x, weight)
           and using_hip(x, 'ff1_geglu_fwd'))
     if ok:
         from .hip_autograd import hip_ff1_geglu

@@ -1,5 +1,6 @@
 """torch.autograd.Function wrappers around the gfx950 HIP kernels."""
 import torch
+from torch.amp import custom_bwd, custom_fwd
 
 from .dispatch import _load_ext
 
@@ -60,6 +61,7 @@ class _LinearFn(torch.autograd.Function):
     K = M reduction is a different shape class)."""
 
     @staticmethod
+    @custom_fwd(device_type='cuda', cast_inputs=torch.bfloat16)
     def forward(ctx, x, weight, bias, residual):
         ext = _load_ext()
         x = x.contiguous()
@@ -72,6 +74,7 @@ class _LinearFn(torch.autograd.Function):
             residual.contiguous() if residual is not None else None)
 
     @staticmethod
+    @custom_bwd(device_type='cuda')
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
         dy2 = dy.contiguous().reshape(-1, dy.shape[-1])
@@ -98,6 +101,7 @@ class _FF1GegluFn(torch.autograd.Function):
     output)."""
 
     @staticmethod
+    @custom_fwd(device_type='cuda', cast_inputs=torch.bfloat16)
     def forward(ctx, x, weight, bias):
         ext = _load_ext()
         x = x.contiguous()
@@ -109,6 +113,7 @@ class _FF1GegluFn(torch.autograd.Function):
         return out
 
     @staticmethod
+    @custom_bwd(device_type='cuda')
     def backward(ctx, dy):
         ext = _load_ext()
         x, weight, inter = ctx.saved_tensors

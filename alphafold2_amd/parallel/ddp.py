@@ -29,6 +29,15 @@ import torch
 import torch.distributed as dist
 
 
+def _force_dist():
+    """AF2AMD_FORCE_DIST=1: initialize a process group and run the full
+    DDP machinery even at world_size 1.  This is the single-GPU
+    rehearsal mode for the multi-GPU path — it exercises RCCL init,
+    broadcast, bucketed async all-reduce and finalize ordering (incl.
+    under hipGraph capture) without needing a multi-GPU lease."""
+    return os.environ.get('AF2AMD_FORCE_DIST') == '1'
+
+
 def init_distributed(backend: Optional[str] = None):
     """Initialize torch.distributed from torchrun env vars; no-op when
     single-process.  Returns (rank, world_size, local_rank)."""
@@ -36,14 +45,17 @@ def init_distributed(backend: Optional[str] = None):
         return dist.get_rank(), dist.get_world_size(), \
             int(os.environ.get('LOCAL_RANK', 0))
     world_size = int(os.environ.get('WORLD_SIZE', '1'))
-    if world_size <= 1:
+    if world_size <= 1 and not _force_dist():
         return 0, 1, 0
     if backend is None:
         backend = 'nccl' if torch.cuda.is_available() else 'gloo'
     local_rank = int(os.environ.get('LOCAL_RANK', '0'))
     if backend == 'nccl':
         torch.cuda.set_device(local_rank)
-    dist.init_process_group(backend=backend)
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    os.environ.setdefault('MASTER_PORT', '29511')
+    dist.init_process_group(backend=backend, world_size=world_size,
+                            rank=int(os.environ.get('RANK', '0')))
     return dist.get_rank(), dist.get_world_size(), local_rank
 
 
@@ -106,7 +118,10 @@ class DataParallelEngine:
         self.model = model
         self.group = process_group
         self.world_size = get_world_size()
-        self.enabled = is_distributed()
+        # forced-dist rehearsal: run the full collective machinery even
+        # at world_size 1 (see _force_dist)
+        self.enabled = is_distributed() or \
+            (dist.is_available() and dist.is_initialized() and _force_dist())
         self.grad_dtype = grad_dtype
         # skip_unused_buckets=True suppresses the all-reduce of buckets
         # in which NO param has a grad.  Only safe when every rank runs

@@ -69,7 +69,7 @@ def main():
     else:
         device = torch.device('cpu')
 
-    is_dist = world_size > 1
+    is_dist = world_size > 1 or os.environ.get('AF2AMD_FORCE_DIST') == '1'
     dist = torch.distributed if is_dist else None
 
     torch.manual_seed(1234 + rank)
@@ -89,10 +89,13 @@ def main():
     engine = DataParallelEngine(model, bucket_cap_mb=64)
 
     # hipGraph capture of the whole step removes the launch-bound host
-    # path (~260 ms/step at this config); capturable Adam required
+    # path (~260 ms/step at this config); capturable Adam required.
+    # DDP steps capture too (RCCL collectives are hipGraph-capturable;
+    # validated via the AF2AMD_FORCE_DIST single-GPU rehearsal) —
+    # AF2AMD_GRAPH_DDP=0 opts out.
     use_graph = (device.type == 'cuda' and not args.no_graph
-                 and (world_size == 1
-                      or os.environ.get('AF2AMD_GRAPH_DDP') == '1'))
+                 and (not is_dist
+                      or os.environ.get('AF2AMD_GRAPH_DDP', '1') == '1'))
     try:
         # single fused multi-tensor Adam kernel (ROCm-supported)
         optimizer = torch.optim.Adam(model.parameters(), lr=3e-4,

@@ -153,3 +153,30 @@ def test_overfit_single_batch_gpu_bf16():
         losses.append(float(loss))
     torch.cuda.synchronize()
     assert losses[-1] < losses[0] * 0.7, (losses[0], losses[-1])
+
+
+@pytest.mark.timeout(420)
+def test_ddp_rccl_graph_rehearsal():
+    """Single-GPU rehearsal of the multi-GPU driver path: a 1-rank RCCL
+    process group with the full DDP engine (broadcast, bucketed async
+    all-reduce, finalize) captured inside a hipGraph — validates the
+    AF2AMD_GRAPH_DDP=1 default without a multi-GPU lease."""
+    import json
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env['AF2AMD_FORCE_DIST'] = '1'
+    env['MASTER_ADDR'] = '127.0.0.1'
+    env['MASTER_PORT'] = '29517'
+    out = subprocess.run(
+        [sys.executable, 'bench.py', '--dim', '64', '--depth', '2',
+         '--crop-len', '64', '--msa-depth', '16', '--batch', '1',
+         '--steps', '3', '--warmup', '2'],
+        cwd=root, capture_output=True, text=True, timeout=400, env=env)
+    assert out.returncode == 0, out.stderr[-3000:]
+    assert '# hipGraph capture: True' in out.stdout, out.stdout[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith('{')][-1]
+    d = json.loads(line)
+    assert d['value'] > 0

@@ -4,6 +4,11 @@ msa 128, batch 5).  Run on a GPU box:
 
     python tools/ffgemm_bench.py
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 from alphafold2_amd.ops.dispatch import _load_ext
