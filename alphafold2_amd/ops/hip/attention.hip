@@ -1080,7 +1080,15 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   const bool has_mask = mask.has_value();
   at::Tensor bias_c, mask_u8;
   if (has_bias) bias_c = bias->contiguous();
-  if (has_mask) mask_u8 = mask->to(at::kByte).contiguous();
+  if (has_mask) {
+    // the kernel indexes mask[batch * Lk + j]: a broadcast (1, Lk) mask
+    // at B > 1 would read out of bounds (advisory r01)
+    TORCH_CHECK(mask->dim() == 2 && mask->size(0) == B &&
+                mask->size(1) == Lk,
+                "attn_fwd: mask must be (B, Lk); expand broadcast masks "
+                "host-side");
+    mask_u8 = mask->to(at::kByte).contiguous();
+  }
 
   dim3 grid((Lq + FBQ - 1) / FBQ, B * H);
   auto stream = at::cuda::getCurrentHIPStream();
@@ -1140,7 +1148,13 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   const bool has_mask = mask.has_value();
   at::Tensor bias_c, mask_u8;
   if (has_bias) bias_c = bias->contiguous();
-  if (has_mask) mask_u8 = mask->to(at::kByte).contiguous();
+  if (has_mask) {
+    TORCH_CHECK(mask->dim() == 2 && mask->size(0) == B &&
+                mask->size(1) == Lk,
+                "attn_bwd: mask must be (B, Lk); expand broadcast masks "
+                "host-side");
+    mask_u8 = mask->to(at::kByte).contiguous();
+  }
 
   auto stream = at::cuda::getCurrentHIPStream();
   TView qv = make_view(q), kvv = make_view(k), vv = make_view(v);

@@ -22,9 +22,10 @@ std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x, at::Tensor g,
                                     c10::optional<at::Tensor> rowmask);
 at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
                       c10::optional<at::Tensor> bias,
-                      c10::optional<at::Tensor> resid);
+                      c10::optional<at::Tensor> resid, long stage);
 std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
-                                      c10::optional<at::Tensor> bias);
+                                      c10::optional<at::Tensor> bias,
+                                      long stage);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
@@ -56,10 +57,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_fwd", &linear_fwd,
         "tall-M small-K linear GEMM, bias/residual epilogue (gfx950 MFMA)",
         py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt,
-        py::arg("resid") = c10::nullopt);
+        py::arg("resid") = c10::nullopt, py::arg("stage") = -1);
   m.def("ff1_geglu_fwd", &ff1_geglu_fwd,
         "linear GEMM with fused GEGLU epilogue (gfx950 MFMA)",
-        py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt);
+        py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt,
+        py::arg("stage") = -1);
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (gfx950)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
         py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"));

@@ -37,6 +37,33 @@ constexpr int NT = 256;        // 4 waves
 
 enum Epi { EPI_NONE = 0, EPI_GEGLU = 1, EPI_RESID = 2 };
 
+// STAGE 0: manual reg staging into a +8-padded LDS row (2-way conflicts,
+// VALU-heavy).  STAGE 1: 16-byte global_load_lds into a LINEAR [128][64]
+// tile with the st_16x32 XOR swizzle applied to BOTH the global source
+// mapping and the ds_read address (guide §5: swizzle must be
+// both-sides-or-neither; the involution is byte ^= ((byte>>9)&1)<<5).
+enum Stage { STAGE_PAD = 0, STAGE_GLDS = 1 };
+
+__device__ __forceinline__ int swz(int byte) {
+  return byte ^ (((byte >> 9) & 1) << 5);
+}
+
+template <int STAGE>
+__device__ __forceinline__ int lds_off(int row, int k) {
+  // byte offset of element (row, k) in one staged tile
+  if (STAGE == STAGE_PAD) return (row * PR + k) * 2;
+  return swz((row << 7) + (k << 1));
+}
+
+typedef __attribute__((address_space(3))) void lds_void;
+typedef const __attribute__((address_space(1))) void g_void;
+
+__device__ __forceinline__ void gload_lds16(const pb16* gsrc, char* lds,
+                                            int lds_byte) {
+  __builtin_amdgcn_global_load_lds(
+      (g_void*)gsrc, (lds_void*)(lds + lds_byte), 16, 0, 0);
+}
+
 // bijective XCD swizzle (guide m204): phys blockIdx -> logical work id
 // such that each XCD owns a contiguous logical range.
 __device__ __forceinline__ long xcd_logical(long phys, long nwg) {
@@ -46,7 +73,7 @@ __device__ __forceinline__ long xcd_logical(long phys, long nwg) {
   return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
 }
 
-template <int EPI, bool HAS_BIAS>
+template <int EPI, bool HAS_BIAS, int STAGE = STAGE_PAD>
 __global__ __launch_bounds__(NT, 2)
 void linear_gemm_kernel(const pb16* __restrict__ A,
                         const pb16* __restrict__ W,
@@ -58,8 +85,10 @@ void linear_gemm_kernel(const pb16* __restrict__ A,
   // staging pool: A[128][72] + B[128][72] bf16 = 36 KB; the C restage
   // ([128][128] bf16 = 32 KB) reuses it after the K loop
   __shared__ pb16 pool[2 * BM * PR];
+  char* a_lds_b = reinterpret_cast<char*>(pool);
+  char* b_lds_b = a_lds_b + (STAGE == STAGE_PAD ? BM * PR * 2 : BM * BK * 2);
   pb16* a_lds = pool;
-  pb16* b_lds = pool + BM * PR;
+  pb16* b_lds = reinterpret_cast<pb16*>(b_lds_b);
   pb16* c_lds = pool;
 
   const long nwg = (long)mtiles * ntiles;
@@ -85,36 +114,57 @@ void linear_gemm_kernel(const pb16* __restrict__ A,
 #pragma unroll
     for (int nt = 0; nt < 8; ++nt) acc[mt][nt] = f32x4_t{0, 0, 0, 0};
 
+  // per-residue row mapping for staged B
+  auto b_src_row = [&](int row) {
+    if (EPI == EPI_GEGLU)
+      return row < 64 ? n0 + row : half + n0 + (row - 64);
+    return n0 + row;
+  };
+  auto b_row_ok = [&](int row) {
+    if (EPI == EPI_GEGLU) return (row & 63) < n_cols;
+    return row < n_cols;
+  };
+
   for (int k0 = 0; k0 < K; k0 += BK) {
     const int k_rows = min(BK, K - k0);
     __syncthreads();
-    // stage A and B tiles: 128 rows x 64 k each as 8-element chunks;
-    // 1024 chunks per tile = 4 per thread
+    if (STAGE == STAGE_PAD) {
+      // stage A and B tiles: 128 rows x 64 k each as 8-element chunks;
+      // 1024 chunks per tile = 4 per thread
 #pragma unroll
-    for (int pass = 0; pass < 4; ++pass) {
-      const int idx = threadIdx.x + pass * NT;
-      const int row = idx >> 3;
-      const int kc = (idx & 7) << 3;
-      bf16x8_t av = {};
-      if (row < m_rows && kc < k_rows)
-        av = *reinterpret_cast<const bf16x8_t*>(
-            A + (long)(m0 + row) * K + k0 + kc);
-      *reinterpret_cast<bf16x8_t*>(a_lds + row * PR + kc) = av;
+      for (int pass = 0; pass < 4; ++pass) {
+        const int idx = threadIdx.x + pass * NT;
+        const int row = idx >> 3;
+        const int kc = (idx & 7) << 3;
+        bf16x8_t av = {};
+        if (row < m_rows && kc < k_rows)
+          av = *reinterpret_cast<const bf16x8_t*>(
+              A + (long)(m0 + row) * K + k0 + kc);
+        *reinterpret_cast<bf16x8_t*>(a_lds + row * PR + kc) = av;
 
-      bf16x8_t bv = {};
-      int grow;
-      bool ok;
-      if (EPI == EPI_GEGLU) {
-        grow = row < 64 ? n0 + row : half + n0 + (row - 64);
-        ok = (row & 63) < n_cols;
-      } else {
-        grow = n0 + row;
-        ok = row < n_cols;
+        bf16x8_t bv = {};
+        if (b_row_ok(row) && kc < k_rows)
+          bv = *reinterpret_cast<const bf16x8_t*>(
+              W + (long)b_src_row(row) * K + k0 + kc);
+        *reinterpret_cast<bf16x8_t*>(b_lds + row * PR + kc) = bv;
       }
-      if (ok && kc < k_rows)
-        bv = *reinterpret_cast<const bf16x8_t*>(
-            W + (long)grow * K + k0 + kc);
-      *reinterpret_cast<bf16x8_t*>(b_lds + row * PR + kc) = bv;
+    } else {
+      // async DMA staging: each wave fills 4 x 1024 B chunks per tile;
+      // lane writes dest byte chunk*1024 + lane*16, whose swizzled
+      // logical home decides the global source.  OOB rows are clamped
+      // (garbage values only reach outputs the epilogue guards off);
+      // requires K % 64 == 0 (host-checked).
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int d = (wave * 4 + i) * 1024 + lane * 16;
+        const int lb = swz(d);
+        const int row = lb >> 7;
+        const int kb = (lb & 127) >> 1;
+        const int ar = min(m0 + row, M - 1);
+        gload_lds16(A + (long)ar * K + k0 + kb, a_lds_b, d);
+        const int br = min(b_src_row(row), (EPI == EPI_GEGLU ? N : N) - 1);
+        gload_lds16(W + (long)br * K + k0 + kb, b_lds_b, d);
+      }
     }
     __syncthreads();
 
@@ -123,13 +173,13 @@ void linear_gemm_kernel(const pb16* __restrict__ A,
     for (int kk = 0; kk < 2; ++kk) {
       const int koff = kk * 32 + ((lane >> 4) << 3);
       bf16x8_t af0 = *reinterpret_cast<const bf16x8_t*>(
-          a_lds + (wr + (lane & 15)) * PR + koff);
+          a_lds_b + lds_off<STAGE>(wr + (lane & 15), koff));
       bf16x8_t af1 = *reinterpret_cast<const bf16x8_t*>(
-          a_lds + (wr + 16 + (lane & 15)) * PR + koff);
+          a_lds_b + lds_off<STAGE>(wr + 16 + (lane & 15), koff));
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         bf16x8_t bf = *reinterpret_cast<const bf16x8_t*>(
-            b_lds + (nt * 16 + (lane & 15)) * PR + koff);
+            b_lds_b + lds_off<STAGE>(nt * 16 + (lane & 15), koff));
         acc[0][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af0, bf, acc[0][nt], 0, 0, 0);
         acc[1][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -229,9 +279,10 @@ void linear_gemm_kernel(const pb16* __restrict__ A,
 }  // namespace
 
 // out = x @ W^T (+ bias) (+ residual); x (M, K), W (N, K) — bf16.
+// stage: -1 auto (async DMA staging when K % 64 == 0), else force 0/1.
 at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
                       c10::optional<at::Tensor> bias,
-                      c10::optional<at::Tensor> resid) {
+                      c10::optional<at::Tensor> resid, long stage) {
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
               W.scalar_type() == at::kBFloat16, "linear_fwd: bf16 only");
   TORCH_CHECK(x.is_contiguous() && W.is_contiguous(),
@@ -259,17 +310,23 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
   const pb16* rp = resid.has_value()
       ? reinterpret_cast<const pb16*>(resid->data_ptr()) : nullptr;
 
-#define LAUNCH(EPI, HB)                                                  \
-  hipLaunchKernelGGL((linear_gemm_kernel<EPI, HB>), grid, dim3(NT), 0,   \
-                     stream, reinterpret_cast<const pb16*>(x.data_ptr()),\
+  const bool glds = stage == 1 || (stage < 0 && K % 64 == 0);
+#define LAUNCH(EPI, HB, ST)                                              \
+  hipLaunchKernelGGL((linear_gemm_kernel<EPI, HB, ST>), grid, dim3(NT),  \
+                     0, stream,                                          \
+                     reinterpret_cast<const pb16*>(x.data_ptr()),        \
                      reinterpret_cast<const pb16*>(W.data_ptr()), bp, rp,\
                      reinterpret_cast<pb16*>(out.data_ptr()), nullptr,   \
                      (int)M, N, K, mtiles, ntiles)
+#define LAUNCH_ST(EPI, HB)                                               \
+  do { if (glds) LAUNCH(EPI, HB, STAGE_GLDS);                            \
+       else LAUNCH(EPI, HB, STAGE_PAD); } while (0)
   if (resid.has_value()) {
-    if (bp) LAUNCH(EPI_RESID, true); else LAUNCH(EPI_RESID, false);
+    if (bp) LAUNCH_ST(EPI_RESID, true); else LAUNCH_ST(EPI_RESID, false);
   } else {
-    if (bp) LAUNCH(EPI_NONE, true); else LAUNCH(EPI_NONE, false);
+    if (bp) LAUNCH_ST(EPI_NONE, true); else LAUNCH_ST(EPI_NONE, false);
   }
+#undef LAUNCH_ST
 #undef LAUNCH
   return out;
 }
@@ -277,7 +334,8 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
 // GEGLU-fused FF1: returns (out (…, N/2) = a * gelu(g), inter (…, N))
 // where inter = x @ W^T + bias is the raw pre-activation (for backward).
 std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
-                                      c10::optional<at::Tensor> bias) {
+                                      c10::optional<at::Tensor> bias,
+                                      long stage) {
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
               W.scalar_type() == at::kBFloat16, "ff1_geglu_fwd: bf16 only");
   TORCH_CHECK(x.is_contiguous() && W.is_contiguous(),
@@ -302,15 +360,20 @@ std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
   const pb16* bp = bias.has_value()
       ? reinterpret_cast<const pb16*>(bias->data_ptr()) : nullptr;
 
-#define LAUNCH(HB)                                                       \
-  hipLaunchKernelGGL((linear_gemm_kernel<EPI_GEGLU, HB>), grid,          \
+  const bool glds = stage == 1 || (stage < 0 && K % 64 == 0);
+#define LAUNCH(HB, ST)                                                   \
+  hipLaunchKernelGGL((linear_gemm_kernel<EPI_GEGLU, HB, ST>), grid,      \
                      dim3(NT), 0, stream,                                \
                      reinterpret_cast<const pb16*>(x.data_ptr()),        \
                      reinterpret_cast<const pb16*>(W.data_ptr()), bp,    \
                      nullptr, reinterpret_cast<pb16*>(out.data_ptr()),   \
                      reinterpret_cast<pb16*>(inter.data_ptr()),          \
                      (int)M, N, K, mtiles, ntiles)
-  if (bp) LAUNCH(true); else LAUNCH(false);
+#define LAUNCH_ST(HB)                                                    \
+  do { if (glds) LAUNCH(HB, STAGE_GLDS); else LAUNCH(HB, STAGE_PAD); }   \
+  while (0)
+  if (bp) LAUNCH_ST(true); else LAUNCH_ST(false);
+#undef LAUNCH_ST
 #undef LAUNCH
   return {out, inter};
 }

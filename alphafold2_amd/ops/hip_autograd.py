@@ -46,12 +46,9 @@ def hip_geglu(x):
 
 
 def _dgrad(dy2, weight):
-    """dX = dY @ W through the custom GEMM (A row-major, weight given
-    as (N, K): dgrad needs (K, N) = W.T, a small contiguous copy)."""
-    ext = _load_ext()
-    wt = weight.t().contiguous()
-    if dy2.shape[-1] % 8 == 0 and wt.shape[0] % 8 == 0:
-        return ext.linear_fwd(dy2, wt, None, None)
+    """dX = dY @ W.  hipBLASLt wins this shape class today (measured
+    2x the custom kernel on dgrad_ff1, profiles/r02_ffgemm_ab.log);
+    revisit when the staging upgrade lands."""
     return dy2 @ weight
 
 
@@ -178,6 +175,10 @@ def hip_attention_core(q, k, v, bias=None, mask=None, context_mask=None,
     key_mask = context_mask if context_mask is not None else mask
     if key_mask is not None:
         key_mask = key_mask.to(torch.uint8)
+        if key_mask.shape[0] != q.shape[0]:
+            # the kernel indexes mask[batch * Lk + j]: broadcast masks
+            # (e.g. the (1, Lk) cross-attention default) expand here
+            key_mask = key_mask.expand(q.shape[0], -1).contiguous()
     scale = q.shape[-1] ** -0.5
     if tie_dim is not None:
         return _AttentionTiedFn.apply(q, k, v, bias, key_mask, tie_dim,

@@ -295,7 +295,8 @@ def test_predict_cli(tmp_path):
 
 
 @pytest.mark.timeout(300)
-def test_predict_cli_a3m(tmp_path):
+def test_predict_cli_a3m_no_checkpoint(tmp_path):
+    """predict.py --a3m without --checkpoint (random-init path)."""
     import subprocess, sys
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     a3m = tmp_path / 'q.a3m'

@@ -56,22 +56,26 @@ def main():
 
         t_blas = time_fn(lambda: torch.nn.functional.linear(x, w, b))
         if epi == 'geglu':
-            t_mine = time_fn(lambda: ext.ff1_geglu_fwd(x, w, b))
+            t_pad = time_fn(lambda: ext.ff1_geglu_fwd(x, w, b, 0))
+            t_glds = time_fn(lambda: ext.ff1_geglu_fwd(x, w, b, 1))
             # reference composition cost: linear + geglu
             t_ref2 = time_fn(lambda: ext.geglu_fwd(
                 torch.nn.functional.linear(x, w, b)))
             extra = f'  blas+geglu={t_ref2:.3f}ms'
         elif epi == 'resid':
-            t_mine = time_fn(lambda: ext.linear_fwd(x, w, b, r))
+            t_pad = time_fn(lambda: ext.linear_fwd(x, w, b, r, 0))
+            t_glds = time_fn(lambda: ext.linear_fwd(x, w, b, r, 1))
             t_ref2 = time_fn(
                 lambda: torch.nn.functional.linear(x, w, b) + r)
             extra = f'  blas+add={t_ref2:.3f}ms'
         else:
-            t_mine = time_fn(lambda: ext.linear_fwd(x, w, b, None))
+            t_pad = time_fn(lambda: ext.linear_fwd(x, w, b, None, 0))
+            t_glds = time_fn(lambda: ext.linear_fwd(x, w, b, None, 1))
             extra = ''
         print(f'{name:16s} M={M:7d} K={K:4d} N={N:4d}  '
               f'blas={t_blas:.3f}ms ({fl / t_blas / 1e9:.0f} TF)  '
-              f'mine={t_mine:.3f}ms ({fl / t_mine / 1e9:.0f} TF)'
+              f'pad={t_pad:.3f}ms ({fl / t_pad / 1e9:.0f} TF)  '
+              f'glds={t_glds:.3f}ms ({fl / t_glds / 1e9:.0f} TF)'
               f'{extra}', flush=True)
 
 
