@@ -59,3 +59,47 @@ def load_checkpoint(path, model, optimizer=None, scheduler=None,
             except RuntimeError:
                 pass  # different device count than at save time
     return state.get('step')
+
+
+# ---------------------------------------------------------------------------
+# standard <-> reversible trunk key mapping
+
+# the two trunk engines hold identical sub-modules under different paths:
+#   standard:   net.layers.{i}.layer.{0,1,2,3}.<rest>
+#               (0=pairwise attn block, 1=pair FF, 2=msa attn block,
+#                3=msa FF — models/evoformer.py EvoformerBlock)
+#   reversible: net.blocks.{i}.{pair_attn,pair_ff,msa_attn,msa_ff}.net.<rest>
+#               (models/reversible.py ReversibleEvoformerBlock)
+_REV_SLOT = {'0': 'pair_attn', '1': 'pair_ff', '2': 'msa_attn',
+             '3': 'msa_ff'}
+_STD_SLOT = {v: k for k, v in _REV_SLOT.items()}
+
+
+def convert_trunk_state_dict(state_dict, to):
+    """Rewrite trunk keys between the standard and reversible layouts.
+
+    `to` is 'reversible' or 'standard'.  Non-trunk keys pass through
+    unchanged, so a checkpoint pretrained with one trunk engine can be
+    resumed with the other (the 288 GB sizing lever — VERDICT r01)."""
+    import re
+    assert to in ('reversible', 'standard')
+    out = {}
+    if to == 'reversible':
+        pat = re.compile(r'^net\.layers\.(\d+)\.layer\.([0-3])\.(.+)$')
+        for k, v in state_dict.items():
+            m = pat.match(k)
+            if m:
+                k = (f'net.blocks.{m.group(1)}.{_REV_SLOT[m.group(2)]}'
+                     f'.net.{m.group(3)}')
+            out[k] = v
+    else:
+        pat = re.compile(
+            r'^net\.blocks\.(\d+)\.(pair_attn|pair_ff|msa_attn|msa_ff)'
+            r'\.net\.(.+)$')
+        for k, v in state_dict.items():
+            m = pat.match(k)
+            if m:
+                k = (f'net.layers.{m.group(1)}.layer.'
+                     f'{_STD_SLOT[m.group(2)]}.{m.group(3)}')
+            out[k] = v
+    return out
