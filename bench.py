@@ -37,6 +37,15 @@ def parse_args():
                    choices=['bf16', 'fp32'])
     p.add_argument('--reversible', action='store_true',
                    help='use the reversible trunk execution mode')
+    p.add_argument('--structure-module', type=str, default='none',
+                   choices=['none', 'ipa', 'se3', 'egnn'],
+                   help='BASELINE configs 4-5: train with coordinate '
+                        'prediction through the given structure module '
+                        '(adds a distance-matrix coord loss)')
+    p.add_argument('--predict-angles', action='store_true',
+                   help='add the theta/phi/omega anglegram heads + loss')
+    p.add_argument('--structure-depth', type=int, default=4,
+                   help='structure module refinement iterations')
     p.add_argument('--no-graph', action='store_true',
                    help='disable hipGraph step capture')
     p.add_argument('--checkpoint-blocks', action='store_true',
@@ -74,6 +83,7 @@ def main():
 
     torch.manual_seed(1234 + rank)
 
+    want_coords = args.structure_module != 'none'
     model = Alphafold2(
         dim=args.dim,
         depth=args.depth,
@@ -83,6 +93,11 @@ def main():
         reversible=args.reversible,
         checkpoint_blocks=('ff' if args.checkpoint_ffs
                            else args.checkpoint_blocks),
+        predict_coords=want_coords,
+        predict_angles=args.predict_angles,
+        structure_module_type=(args.structure_module if want_coords
+                               else 'ipa'),
+        structure_module_depth=args.structure_depth,
     ).to(device)
     model.train()
 
@@ -111,6 +126,19 @@ def main():
     seq, msa = batch['seq'], batch['msa']
     mask, msa_mask = batch['mask'], batch['msa_mask']
     target = get_bucketed_distance_matrix(batch['coords'], mask)
+    coords_target = batch['coords']
+    target_dmat = torch.cdist(coords_target, coords_target) \
+        if want_coords else None
+    if args.predict_angles:
+        from alphafold2_amd import constants
+        gen = torch.Generator(device='cpu').manual_seed(24 + rank)
+        angle_targets = {
+            name: torch.randint(0, buckets,
+                                (args.batch, args.crop_len, args.crop_len),
+                                generator=gen).to(device)
+            for name, buckets in (('theta', constants.THETA_BUCKETS),
+                                  ('phi', constants.PHI_BUCKETS),
+                                  ('omega', constants.OMEGA_BUCKETS))}
 
     def step():
         optimizer.zero_grad(set_to_none=not use_graph)
@@ -121,12 +149,28 @@ def main():
             import contextlib
             ctx = contextlib.nullcontext()
         with ctx:
-            ret = model(seq, msa, mask=mask, msa_mask=msa_mask)
+            if want_coords:
+                coords, ret = model(seq, msa, mask=mask, msa_mask=msa_mask,
+                                    return_aux_logits=True)
+            else:
+                ret = model(seq, msa, mask=mask, msa_mask=msa_mask)
             logits = ret.distance.permute(0, 3, 1, 2)
             loss = torch.nn.functional.cross_entropy(
                 logits.float(), target, ignore_index=-100)
             if ret.msa_mlm_loss is not None:
                 loss = loss + ret.msa_mlm_loss.float()
+            if want_coords:
+                # alignment-free coordinate loss (distance-matrix MSE,
+                # reference train_end2end intent) — graph-capturable,
+                # no host-synced Kabsch in the hot loop
+                pred_dmat = torch.cdist(coords.float(), coords.float())
+                loss = loss + torch.nn.functional.smooth_l1_loss(
+                    pred_dmat, target_dmat)
+            if args.predict_angles:
+                for name, tgt in angle_targets.items():
+                    alog = getattr(ret, f'{name}_logits')
+                    loss = loss + 0.1 * torch.nn.functional.cross_entropy(
+                        alog.permute(0, 3, 1, 2).float(), tgt)
         loss.backward()
         engine.finalize()
         optimizer.step()
@@ -184,7 +228,10 @@ def main():
             "config": {
                 "model": f"alphafold2 evoformer dim={args.dim} depth={args.depth} "
                          f"heads={args.heads} dim_head={args.dim_head}"
-                         + (" reversible" if args.reversible else ""),
+                         + (" reversible" if args.reversible else "")
+                         + (f" structure={args.structure_module}"
+                            f"x{args.structure_depth}" if want_coords else "")
+                         + (" angles" if args.predict_angles else ""),
                 "global_batch": args.batch * world_size,
                 "seq_len": args.crop_len,
                 "msa_depth": args.msa_depth,
