@@ -1040,6 +1040,379 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
   }
 }
 
+// ---------------------------------------------------------------------------
+// split backward (r02): the combined dK/dV kernel sits at 196-249 VGPRs
+// = 2 waves/SIMD.  Splitting recomputes S^T/P^T once more (5 -> 7
+// GEMM-equivalents) but the dV pass drops to ~4 waves/SIMD and the dK
+// pass to 3 — at single-digit MfmaUtil the residency win dominates.
+
+template <bool HAS_BIAS, bool HAS_MASK>
+__global__ __launch_bounds__(256, 4)
+void attn_bwd_dv_kernel(TView q, TView k,
+                        const bf16_t* __restrict__ bias,
+                        const unsigned char* __restrict__ mask,
+                        TView dout,
+                        const float* __restrict__ lse,
+                        TViewMut dv,
+                        int Lq, int Lk, int heads, int bias_repeat,
+                        float scale) {
+  __shared__ char k_lds[BK * ROWB];
+  __shared__ char q_lds[BQ * ROWB];
+  __shared__ char dot_lds[BQ * ROWB];  // dO transposed: [dv][q]
+  // bias tile [q][kv] and the P^T scratch ALIAS: the bias is fully
+  // consumed (P^T formed in regs) before the scratch write, and the
+  // barrier below separates the cross-wave read/write windows.  Saves
+  // 8 KB -> 4 workgroups/CU.
+  __shared__ char sb_lds[BQ * ROWB];
+  char* b_lds = sb_lds;
+  char (*s_lds)[16 * ROWB] =
+      reinterpret_cast<char (*)[16 * ROWB]>(sb_lds);
+  __shared__ float lse_lds[BQ];
+  __shared__ unsigned char m_lds[BK];
+
+  const int ktile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int batch = bh / heads;
+  const int head = bh - batch * heads;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  const bf16_t* k_g = k.base(batch, head) + (long)ktile * BK * k.rs;
+  const bf16_t* q_g = q.base(batch, head);
+  const bf16_t* do_g = dout.base(batch, head);
+  const float* lse_g = lse + (long)bh * Lq;
+  const bf16_t* bias_g = HAS_BIAS
+      ? bias + (long)((batch / bias_repeat) * heads + head) * Lq * Lk
+      : nullptr;
+
+  const int kv_rows = min(BK, Lk - ktile * BK);
+  stage_tile(k_g, k.rs, kv_rows, k_lds);
+  if (HAS_MASK && threadIdx.x < BK) {
+    m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
+        ? mask[(long)batch * Lk + ktile * BK + threadIdx.x] : 0;
+  }
+  __syncthreads();
+
+  bf16x8 k_frag[2];
+#pragma unroll
+  for (int dblk = 0; dblk < 2; ++dblk)
+    k_frag[dblk] = frag_row(k_lds, wave * 16 + (lane & 15), dblk);
+  bool krow_ok_c[4];
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = wave * 16 + (lane >> 4) * 4 + reg;
+    krow_ok_c[reg] = row < kv_rows && (!HAS_MASK || m_lds[row]);
+  }
+
+  f32x4 dv_acc[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c) dv_acc[c] = f32x4{0, 0, 0, 0};
+
+  const int n_q = (Lq + BQ - 1) / BQ;
+  // direct (prefetch-free) staging: this kernel trades the async-stage
+  // register cost for residency — occupancy is its lever
+  for (int t = 0; t < n_q; ++t) {
+    const int q_rows = min(BQ, Lq - t * BQ);
+    __syncthreads();
+    stage_tile(q_g + (long)t * BQ * q.rs, q.rs, q_rows, q_lds);
+    stage_tile_t(do_g + (long)t * BQ * dout.rs, dout.rs, q_rows, dot_lds);
+    if (HAS_BIAS) {
+      stage_tile_rowstride(bias_g + (long)t * BQ * Lk + (long)ktile * BK,
+                           Lk, q_rows, min(BK, Lk - ktile * BK), b_lds);
+    }
+    if (threadIdx.x < BQ) {
+      const int qq = t * BQ + threadIdx.x;
+      lse_lds[threadIdx.x] = (threadIdx.x < q_rows) ? lse_g[qq] : NEG_INF;
+    }
+    __syncthreads();
+
+    // S^T = K Q^T ; P^T = exp(S^T*scale + bias^T - lse[q])
+    f32x4 pt[4];
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      f32x4 acc = {0, 0, 0, 0};
+#pragma unroll
+      for (int dblk = 0; dblk < 2; ++dblk) {
+        bf16x8 qf = frag_row(q_lds, c * 16 + (lane & 15), dblk);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[dblk], qf, acc,
+                                                      0, 0, 0);
+      }
+      pt[c] = acc;
+    }
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int qcol = c * 16 + (lane & 15);
+      const float l = lse_lds[qcol];
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int kvrow = wave * 16 + (lane >> 4) * 4 + reg;
+        float val = pt[c][reg] * scale;
+        if (HAS_BIAS && krow_ok_c[reg] && qcol < q_rows)
+          val += to_f32(*reinterpret_cast<const bf16_t*>(
+              b_lds + swz(qcol, kvrow * (int)sizeof(bf16_t))));
+        pt[c][reg] = (krow_ok_c[reg] && qcol < q_rows && l > NEG_INF)
+            ? __expf(val - l) : 0.f;
+      }
+    }
+
+    // dV += P^T dO : A = P^T (k = q, via per-wave LDS scratch).
+    // The scratch aliases the bias tile: wait until EVERY wave has
+    // finished its bias reads before overwriting.
+    if (HAS_BIAS) __syncthreads();
+    char* sw = s_lds[wave];
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int col = c * 16 + (lane & 15);
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = (lane >> 4) * 4 + reg;
+        *reinterpret_cast<bf16_t*>(sw + swz(row, col * (int)sizeof(bf16_t)))
+            = (bf16_t)pt[c][reg];
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    bf16x8 pt_frag[2];
+#pragma unroll
+    for (int kblk = 0; kblk < 2; ++kblk)
+      pt_frag[kblk] = frag_row(sw, lane & 15, kblk);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      f32x4 acc = dv_acc[c];
+#pragma unroll
+      for (int kblk = 0; kblk < 2; ++kblk) {
+        bf16x8 dof = frag_row(dot_lds, c * 16 + (lane & 15), kblk);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag[kblk], dof, acc,
+                                                      0, 0, 0);
+      }
+      dv_acc[c] = acc;
+    }
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  bf16_t* dv_g = dv.base(batch, head) + (long)ktile * BK * dv.rs;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = wave * 16 + (lane >> 4) * 4 + reg;
+    if (row < kv_rows) {
+#pragma unroll
+      for (int c = 0; c < 4; ++c)
+        dv_g[(long)row * dv.rs + c * 16 + (lane & 15)] =
+            (bf16_t)dv_acc[c][reg];
+    }
+  }
+}
+
+template <bool HAS_BIAS, bool HAS_MASK, bool NEED_DBIAS>
+__global__ __launch_bounds__(256, 3)
+void attn_bwd_dk_kernel(TView q, TView k, TView v,
+                        const bf16_t* __restrict__ bias,
+                        const unsigned char* __restrict__ mask,
+                        TView dout,
+                        const float* __restrict__ lse,
+                        const float* __restrict__ delta,
+                        TViewMut dk,
+                        float* __restrict__ dbias,
+                        int dbias_chunks, long dbias_stride,
+                        int Lq, int Lk, int heads, int bias_repeat,
+                        float scale) {
+  __shared__ char k_lds[BK * ROWB];
+  __shared__ char v_lds[BK * ROWB];
+  __shared__ char q_lds[BQ * ROWB];
+  __shared__ char qt_lds[BQ * ROWB];   // Q transposed: [d][q]
+  __shared__ char do_lds[BQ * ROWB];
+  // bias tile and dS scratch alias (see dv kernel) — 48 KB total LDS
+  __shared__ char sb_lds[BQ * ROWB];
+  char* b_lds = sb_lds;
+  char (*s_lds)[16 * ROWB] =
+      reinterpret_cast<char (*)[16 * ROWB]>(sb_lds);
+  __shared__ float lse_lds[BQ];
+  __shared__ float delta_lds[BQ];
+  __shared__ unsigned char m_lds[BK];
+
+  const int ktile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int batch = bh / heads;
+  const int head = bh - batch * heads;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  const bf16_t* k_g = k.base(batch, head) + (long)ktile * BK * k.rs;
+  const bf16_t* v_g = v.base(batch, head) + (long)ktile * BK * v.rs;
+  const bf16_t* q_g = q.base(batch, head);
+  const bf16_t* do_g = dout.base(batch, head);
+  const float* lse_g = lse + (long)bh * Lq;
+  const float* delta_g = delta + (long)bh * Lq;
+  const int bias_batch = batch / bias_repeat;
+  const bf16_t* bias_g = HAS_BIAS
+      ? bias + (long)(bias_batch * heads + head) * Lq * Lk : nullptr;
+  float* dbias_g = nullptr;
+  if (NEED_DBIAS) {
+    const int chunk = (batch % bias_repeat) % dbias_chunks;
+    dbias_g = dbias + (long)chunk * dbias_stride
+        + (long)(bias_batch * heads + head) * Lq * Lk;
+  }
+
+  const int kv_rows = min(BK, Lk - ktile * BK);
+  stage_tile(k_g, k.rs, kv_rows, k_lds);
+  stage_tile(v_g, v.rs, kv_rows, v_lds);
+  if (HAS_MASK && threadIdx.x < BK) {
+    m_lds[threadIdx.x] = (threadIdx.x < kv_rows)
+        ? mask[(long)batch * Lk + ktile * BK + threadIdx.x] : 0;
+  }
+  __syncthreads();
+
+  const int krow = wave * 16 + (lane & 15);
+  bf16x8 k_frag[2], v_frag[2];
+#pragma unroll
+  for (int dblk = 0; dblk < 2; ++dblk) {
+    k_frag[dblk] = frag_row(k_lds, krow, dblk);
+    v_frag[dblk] = frag_row(v_lds, krow, dblk);
+  }
+  bool krow_ok_c[4];
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = wave * 16 + (lane >> 4) * 4 + reg;
+    krow_ok_c[reg] = row < kv_rows && (!HAS_MASK || m_lds[row]);
+  }
+
+  f32x4 dk_acc[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c) dk_acc[c] = f32x4{0, 0, 0, 0};
+
+  const int n_q = (Lq + BQ - 1) / BQ;
+  // prefetch-free staging (register diet for residency); the dual
+  // q layouts still come from one StageRegs load
+  StageRegs qreg;
+  for (int t = 0; t < n_q; ++t) {
+    const int q_rows = min(BQ, Lq - t * BQ);
+    stage_load(q_g + (long)t * BQ * q.rs, q.rs, q_rows, qreg);
+    __syncthreads();
+    stage_store(qreg, q_lds);
+    stage_store_t(qreg, qt_lds);
+    stage_tile(do_g + (long)t * BQ * dout.rs, dout.rs, q_rows, do_lds);
+    if (HAS_BIAS) {
+      stage_tile_rowstride(bias_g + (long)t * BQ * Lk + (long)ktile * BK,
+                           Lk, q_rows, min(BK, Lk - ktile * BK), b_lds);
+    }
+    if (threadIdx.x < BQ) {
+      const int qq = t * BQ + threadIdx.x;
+      lse_lds[threadIdx.x] = (threadIdx.x < q_rows) ? lse_g[qq] : NEG_INF;
+      delta_lds[threadIdx.x] = (threadIdx.x < q_rows) ? delta_g[qq] : 0.f;
+    }
+    __syncthreads();
+
+    // S^T = K Q^T ; P^T
+    f32x4 pt[4], dpt[4];
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      f32x4 acc = {0, 0, 0, 0};
+#pragma unroll
+      for (int dblk = 0; dblk < 2; ++dblk) {
+        bf16x8 qf = frag_row(q_lds, c * 16 + (lane & 15), dblk);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[dblk], qf, acc,
+                                                      0, 0, 0);
+      }
+      pt[c] = acc;
+    }
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int qcol = c * 16 + (lane & 15);
+      const float l = lse_lds[qcol];
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int kvrow = wave * 16 + (lane >> 4) * 4 + reg;
+        float val = pt[c][reg] * scale;
+        if (HAS_BIAS && krow_ok_c[reg] && qcol < q_rows)
+          val += to_f32(*reinterpret_cast<const bf16_t*>(
+              b_lds + swz(qcol, kvrow * (int)sizeof(bf16_t))));
+        pt[c][reg] = (krow_ok_c[reg] && qcol < q_rows && l > NEG_INF)
+            ? __expf(val - l) : 0.f;
+      }
+    }
+
+    // dP^T = V dO^T
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      f32x4 acc = {0, 0, 0, 0};
+#pragma unroll
+      for (int dblk = 0; dblk < 2; ++dblk) {
+        bf16x8 dof = frag_row(do_lds, c * 16 + (lane & 15), dblk);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[dblk], dof, acc,
+                                                      0, 0, 0);
+      }
+      dpt[c] = acc;
+    }
+
+    // dS^T = P^T (dP^T - delta[q]) ; stage for dK (+ dBias drain).
+    // the scratch aliases the bias tile — barrier off the bias reads
+    if (HAS_BIAS) __syncthreads();
+    char* sw = s_lds[wave];
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int qcol = c * 16 + (lane & 15);
+      const float dl = delta_lds[qcol];
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float ds_total = pt[c][reg] * (dpt[c][reg] - dl);
+        const int row = (lane >> 4) * 4 + reg;
+        *reinterpret_cast<bf16_t*>(sw + swz(row, qcol * (int)sizeof(bf16_t)))
+            = (bf16_t)ds_total;
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    bf16x8 dst_frag[2];
+#pragma unroll
+    for (int kblk = 0; kblk < 2; ++kblk)
+      dst_frag[kblk] = frag_row(sw, lane & 15, kblk);
+
+    if (NEED_DBIAS) {
+      __syncthreads();
+      const int kvr = lane;
+      // unrolling this drain costs ~30 VGPRs of address computation —
+      // keep it rolled (occupancy is this kernel's lever)
+#pragma clang loop unroll(disable)
+      for (int j = 0; j < 16; ++j) {
+        const int qq = j * NWAVES + wave;
+        if (qq < q_rows && kvr < kv_rows) {
+          float dsv = to_f32(*reinterpret_cast<const bf16_t*>(
+              s_lds[kvr >> 4] + swz(kvr & 15, qq * (int)sizeof(bf16_t))));
+          float* addr = dbias_g + (long)(t * BQ + qq) * Lk
+              + (long)ktile * BK + kvr;
+          if (bias_repeat == 1) *addr = dsv; else atomicAdd(addr, dsv);
+        }
+      }
+    }
+
+    // dK += dS^T Q
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      f32x4 acc = dk_acc[c];
+#pragma unroll
+      for (int kblk = 0; kblk < 2; ++kblk) {
+        bf16x8 qf = frag_row(qt_lds, c * 16 + (lane & 15), kblk);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dst_frag[kblk], qf, acc,
+                                                      0, 0, 0);
+      }
+      dk_acc[c] = acc;
+    }
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  bf16_t* dk_g = dk.base(batch, head) + (long)ktile * BK * dk.rs;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = wave * 16 + (lane >> 4) * 4 + reg;
+    if (row < kv_rows) {
+#pragma unroll
+      for (int c = 0; c < 4; ++c)
+        dk_g[(long)row * dk.rs + c * 16 + (lane & 15)] =
+            (bf16_t)(dk_acc[c][reg] * scale);
+    }
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1197,14 +1570,54 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
       dbias_chunks, need_dbias ? dbias.stride(0) : 0L,                        \
       Lq, Lk, H, (int)bias_repeat, (float)scale)
 
-  if (need_dbias) {
-    if (has_mask) DISPATCH_DKV(true, true, true);
-    else DISPATCH_DKV(true, false, true);
-  } else if (has_bias && has_mask) DISPATCH_DKV(true, true, false);
-  else if (has_bias) DISPATCH_DKV(true, false, false);
-  else if (has_mask) DISPATCH_DKV(false, true, false);
-  else DISPATCH_DKV(false, false, false);
+#define DISPATCH_DV(HB, HM)                                                   \
+  hipLaunchKernelGGL((attn_bwd_dv_kernel<HB, HM>), grid_k, dim3(256), 0,      \
+      stream, qv, kvv,                                                        \
+      has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
+      has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
+      dov, lse.data_ptr<float>(), dvv,                                        \
+      Lq, Lk, H, (int)bias_repeat, (float)scale)
+
+#define DISPATCH_DK(HB, HM, DB)                                               \
+  hipLaunchKernelGGL((attn_bwd_dk_kernel<HB, HM, DB>), grid_k, dim3(256), 0,  \
+      stream, qv, kvv, vv,                                                    \
+      has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
+      has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
+      dov, lse.data_ptr<float>(), delta.data_ptr<float>(), dkv,               \
+      DB ? dbias.data_ptr<float>() : nullptr,                                 \
+      dbias_chunks, need_dbias ? dbias.stride(0) : 0L,                        \
+      Lq, Lk, H, (int)bias_repeat, (float)scale)
+
+  // split dV/dK passes (2 -> 4/3 waves per SIMD) unless opted out
+  static const bool split_dkv = [] {
+    const char* e = getenv("AF2AMD_SPLIT_DKV");
+    return e == nullptr || e[0] != '0';
+  }();
+
+  if (split_dkv) {
+    if (has_bias && has_mask) DISPATCH_DV(true, true);
+    else if (has_bias) DISPATCH_DV(true, false);
+    else if (has_mask) DISPATCH_DV(false, true);
+    else DISPATCH_DV(false, false);
+    if (need_dbias) {
+      if (has_mask) DISPATCH_DK(true, true, true);
+      else DISPATCH_DK(true, false, true);
+    } else if (has_bias && has_mask) DISPATCH_DK(true, true, false);
+    else if (has_bias) DISPATCH_DK(true, false, false);
+    else if (has_mask) DISPATCH_DK(false, true, false);
+    else DISPATCH_DK(false, false, false);
+  } else {
+    if (need_dbias) {
+      if (has_mask) DISPATCH_DKV(true, true, true);
+      else DISPATCH_DKV(true, false, true);
+    } else if (has_bias && has_mask) DISPATCH_DKV(true, true, false);
+    else if (has_bias) DISPATCH_DKV(true, false, false);
+    else if (has_mask) DISPATCH_DKV(false, true, false);
+    else DISPATCH_DKV(false, false, false);
+  }
 #undef DISPATCH_DKV
+#undef DISPATCH_DV
+#undef DISPATCH_DK
 
   std::vector<at::Tensor> ret = {dq, dk, dv};
   if (need_dbias) ret.push_back(dbias.sum(0));
