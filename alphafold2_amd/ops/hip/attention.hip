@@ -1588,10 +1588,13 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
       dbias_chunks, need_dbias ? dbias.stride(0) : 0L,                        \
       Lq, Lk, H, (int)bias_repeat, (float)scale)
 
-  // split dV/dK passes (2 -> 4/3 waves per SIMD) unless opted out
+  // split dV/dK passes (4/3 waves per SIMD, but prefetch-free and
+  // +40% flops): measured SLOWER end-to-end than the combined kernel
+  // (748 vs 699 ms/step, r02) — combined stays default;
+  // AF2AMD_SPLIT_DKV=1 re-enables for tuning.
   static const bool split_dkv = [] {
     const char* e = getenv("AF2AMD_SPLIT_DKV");
-    return e == nullptr || e[0] != '0';
+    return e != nullptr && e[0] == '1';
   }();
 
   if (split_dkv) {

@@ -26,6 +26,7 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
 std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
                                       c10::optional<at::Tensor> bias,
                                       long stage);
+at::Tensor wgrad(at::Tensor dY, at::Tensor X);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
@@ -62,6 +63,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "linear GEMM with fused GEGLU epilogue (gfx950 MFMA)",
         py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt,
         py::arg("stage") = -1);
+  m.def("wgrad", &wgrad,
+        "split-K weight-gradient GEMM dY^T @ X (gfx950 MFMA, fp32 out)",
+        py::arg("dY"), py::arg("X"));
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (gfx950)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
         py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"));

@@ -455,3 +455,16 @@ def test_ff1_geglu_autograd_parity(ext):
                    (b1.grad, b2.grad)]:
         denom = g2.abs().max().item() + 1e-6
         assert (g1.float() - g2).abs().max().item() / denom < 6e-2
+
+
+@pytest.mark.parametrize("K,M,N", [(8192, 2048, 256), (4096, 256, 512),
+                                   (1000, 640, 256), (131072, 256, 256)])
+def test_wgrad_parity(ext, K, M, N):
+    torch.manual_seed(0)
+    dy = torch.randn(K, M, device='cuda', dtype=torch.bfloat16) * 0.5
+    x = torch.randn(K, N, device='cuda', dtype=torch.bfloat16) * 0.5
+    ref = dy.float().t() @ x.float()
+    got = ext.wgrad(dy, x)
+    denom = ref.abs().max().item() + 1e-6
+    rel = (got - ref).abs().max().item() / denom
+    assert rel < 3e-2, rel

@@ -4,6 +4,11 @@ full-bench box noise).  Run on an MI355X:
 
     PYTHONPATH=/root/repo python tools/attn_bench.py
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 from alphafold2_amd.ops.dispatch import _load_ext

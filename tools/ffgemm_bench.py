@@ -78,6 +78,24 @@ def main():
               f'glds={t_glds:.3f}ms ({fl / t_glds / 1e9:.0f} TF)'
               f'{extra}', flush=True)
 
+    print('--- wgrad (dW = dY^T @ X, split-K) ---', flush=True)
+    for name, K, M, N in [
+        ('ff1_wgrad_msa', 5 * 128 * 256, 2048, 256),
+        ('ff1_wgrad_pair', 5 * 256 * 256, 2048, 256),
+        ('ff2_wgrad_pair', 5 * 256 * 256, 256, 1024),
+        ('qkvg_wgrad', 5 * 256 * 256, 2048, 256),
+        ('out_wgrad', 5 * 256 * 256, 256, 512),
+    ]:
+        dy = torch.randn(K, M, device=dev, dtype=dt) * 0.1
+        x = torch.randn(K, N, device=dev, dtype=dt) * 0.1
+        fl = 2.0 * K * M * N
+        t_blas = time_fn(lambda: dy.t() @ x)
+        t_mine = time_fn(lambda: ext.wgrad(dy, x))
+        print(f'{name:16s} K={K:7d} M={M:4d} N={N:4d}  '
+              f'blas={t_blas:.3f}ms ({fl / t_blas / 1e9:.0f} TF)  '
+              f'mine={t_mine:.3f}ms ({fl / t_mine / 1e9:.0f} TF)',
+              flush=True)
+
 
 if __name__ == '__main__':
     main()
