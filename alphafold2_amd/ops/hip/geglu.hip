@@ -12,15 +12,23 @@
 
 namespace {
 
-// row-outer loop: no per-element division (a runtime `idx / H` in the
-// hot loop serializes on the integer-div unit)
+// row-outer loop with multi-row packing: when H/VEC < blockDim the
+// block covers several rows per iteration (shift-based row split for
+// power-of-2 H — a runtime `idx / H` in the hot loop serializes on the
+// integer-div unit).  row_shift = log2(H / VEC); 0 disables packing.
 template <typename T, int VEC>
 __global__ void geglu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
-                                 long rows, int H) {
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+                                 long rows, int H, int row_shift) {
+  const int sub = row_shift ? (threadIdx.x >> row_shift) : 0;
+  const int rows_per_iter = row_shift ? (blockDim.x >> row_shift) : 1;
+  const int col0 = row_shift
+      ? (threadIdx.x & ((1 << row_shift) - 1)) * VEC : threadIdx.x * VEC;
+  const int cstep = row_shift ? H : blockDim.x * VEC;
+  for (long row = (long)blockIdx.x * rows_per_iter + sub; row < rows;
+       row += (long)gridDim.x * rows_per_iter) {
     const T* xr = x + row * (2L * H);
     T* yr = y + row * (long)H;
-    for (int col = threadIdx.x * VEC; col < H; col += blockDim.x * VEC) {
+    for (int col = col0; col < H; col += cstep) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float a = to_f32(xr[col + k]);
@@ -34,12 +42,19 @@ __global__ void geglu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
 template <typename T, int VEC>
 __global__ void geglu_bwd_kernel(const T* __restrict__ dy,
                                  const T* __restrict__ x,
-                                 T* __restrict__ dx, long rows, int H) {
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+                                 T* __restrict__ dx, long rows, int H,
+                                 int row_shift) {
+  const int sub = row_shift ? (threadIdx.x >> row_shift) : 0;
+  const int rows_per_iter = row_shift ? (blockDim.x >> row_shift) : 1;
+  const int col0 = row_shift
+      ? (threadIdx.x & ((1 << row_shift) - 1)) * VEC : threadIdx.x * VEC;
+  const int cstep = row_shift ? H : blockDim.x * VEC;
+  for (long row = (long)blockIdx.x * rows_per_iter + sub; row < rows;
+       row += (long)gridDim.x * rows_per_iter) {
     const T* dyr = dy + row * (long)H;
     const T* xr = x + row * (2L * H);
     T* dxr = dx + row * (2L * H);
-    for (int col = threadIdx.x * VEC; col < H; col += blockDim.x * VEC) {
+    for (int col = col0; col < H; col += cstep) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float a = to_f32(xr[col + k]);
@@ -71,16 +86,23 @@ at::Tensor geglu_fwd(at::Tensor x) {
 
 #define LAUNCH(T, VEC)                                                   \
   do {                                                                   \
-    long grid = rows < 4096 ? rows : 4096;                               \
+    int colt = H / VEC;                                                  \
+    int row_shift = 0;                                                   \
+    if (colt < block && colt > 0 && (colt & (colt - 1)) == 0)            \
+      row_shift = __builtin_ctz(colt);                                   \
+    const int rpi = row_shift ? block >> row_shift : 1;                  \
+    long grid = (rows + rpi - 1) / rpi;                                  \
+    if (grid > 8192) grid = 8192;                                        \
     if (grid < 1) grid = 1;                                              \
     hipLaunchKernelGGL((geglu_fwd_kernel<T, VEC>), dim3(grid),           \
                        dim3(block), 0, stream,                           \
                        reinterpret_cast<const T*>(x.data_ptr()),         \
-                       reinterpret_cast<T*>(y.data_ptr()), rows, H);     \
+                       reinterpret_cast<T*>(y.data_ptr()), rows, H,      \
+                       row_shift);                                       \
   } while (0)
 
-  // adaptive VEC: keep all 256 lanes busy when H < 2048
-  const bool vec8 = (H % 8) == 0 && H >= 2048;
+  // multi-row packing keeps all lanes busy at any H; prefer 16B lanes
+  const bool vec8 = (H % 8) == 0;
   const bool vec4 = (H % 4) == 0;
   if (x.scalar_type() == at::kBFloat16) {
     if (vec8) LAUNCH(__hip_bfloat16, 8);
@@ -113,17 +135,24 @@ at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x) {
 
 #define LAUNCH(T, VEC)                                                   \
   do {                                                                   \
-    long grid = rows < 4096 ? rows : 4096;                               \
+    int colt = H / VEC;                                                  \
+    int row_shift = 0;                                                   \
+    if (colt < block && colt > 0 && (colt & (colt - 1)) == 0)            \
+      row_shift = __builtin_ctz(colt);                                   \
+    const int rpi = row_shift ? block >> row_shift : 1;                  \
+    long grid = (rows + rpi - 1) / rpi;                                  \
+    if (grid > 8192) grid = 8192;                                        \
     if (grid < 1) grid = 1;                                              \
     hipLaunchKernelGGL((geglu_bwd_kernel<T, VEC>), dim3(grid),           \
                        dim3(block), 0, stream,                           \
                        reinterpret_cast<const T*>(dy.data_ptr()),        \
                        reinterpret_cast<const T*>(x.data_ptr()),         \
-                       reinterpret_cast<T*>(dx.data_ptr()), rows, H);    \
+                       reinterpret_cast<T*>(dx.data_ptr()), rows, H,     \
+                       row_shift);                                       \
   } while (0)
 
-  // adaptive VEC: keep all 256 lanes busy when H < 2048
-  const bool vec8 = (H % 8) == 0 && H >= 2048;
+  // multi-row packing keeps all lanes busy at any H; prefer 16B lanes
+  const bool vec8 = (H % 8) == 0;
   const bool vec4 = (H % 4) == 0;
   if (x.scalar_type() == at::kBFloat16) {
     if (vec8) LAUNCH(__hip_bfloat16, 8);
