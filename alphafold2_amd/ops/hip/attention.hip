@@ -362,7 +362,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
                      const bf16_t* __restrict__ bias,
                      const unsigned char* __restrict__ mask,
                      TViewMut out, float* __restrict__ lse,
-                     int Lq, int Lk, int heads, int bias_repeat,
+                     int Lq, int Lk, int heads, int bias_repeat, int q_repeat,
                      float scale) {
   __shared__ char q_lds[FBQ * ROWB];
   // K / V^T / mask tiles are DOUBLE-buffered so the loop needs a single
@@ -382,7 +382,7 @@ void attn_fwd_kernel(TView q, TView k, TView v,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  const bf16_t* q_g = q.base(batch, head) + (long)qtile * FBQ * q.rs;
+  const bf16_t* q_g = q.base(batch / q_repeat, head) + (long)qtile * FBQ * q.rs;
   const bf16_t* k_g = k.base(batch, head);
   const bf16_t* v_g = v.base(batch, head);
   const bf16_t* bias_g = nullptr;
@@ -597,7 +597,7 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
                         const float* __restrict__ lse,
                         const float* __restrict__ delta,
                         TViewMut dq,
-                        int Lq, int Lk, int heads, int bias_repeat,
+                        int Lq, int Lk, int heads, int bias_repeat, int q_repeat,
                         float scale) {
   __shared__ char q_lds[BQ * ROWB];
   __shared__ char do_lds[BQ * ROWB];
@@ -614,7 +614,7 @@ void attn_bwd_dq_kernel(TView q, TView k, TView v,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  const bf16_t* q_g = q.base(batch, head) + (long)qtile * BQ * q.rs;
+  const bf16_t* q_g = q.base(batch / q_repeat, head) + (long)qtile * BQ * q.rs;
   const bf16_t* do_g = dout.base(batch, head) + (long)qtile * BQ * dout.rs;
   const bf16_t* k_g = k.base(batch, head);
   const bf16_t* v_g = v.base(batch, head);
@@ -790,7 +790,7 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
                          TViewMut dk, TViewMut dv,
                          float* __restrict__ dbias,
                          int dbias_chunks, long dbias_stride,
-                         int Lq, int Lk, int heads, int bias_repeat,
+                         int Lq, int Lk, int heads, int bias_repeat, int q_repeat,
                          float scale) {
   __shared__ char k_lds[BK * ROWB];
   __shared__ char v_lds[BK * ROWB];
@@ -813,7 +813,7 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
 
   const bf16_t* k_g = k.base(batch, head) + (long)ktile * BK * k.rs;
   const bf16_t* v_g = v.base(batch, head) + (long)ktile * BK * v.rs;
-  const bf16_t* q_g = q.base(batch, head);
+  const bf16_t* q_g = q.base(batch / q_repeat, head);
   const bf16_t* do_g = dout.base(batch, head);
   const float* lse_g = lse + (long)bh * Lq;
   const float* delta_g = delta + (long)bh * Lq;
@@ -1054,7 +1054,7 @@ void attn_bwd_dv_kernel(TView q, TView k,
                         TView dout,
                         const float* __restrict__ lse,
                         TViewMut dv,
-                        int Lq, int Lk, int heads, int bias_repeat,
+                        int Lq, int Lk, int heads, int bias_repeat, int q_repeat,
                         float scale) {
   __shared__ char k_lds[BK * ROWB];
   __shared__ char q_lds[BQ * ROWB];
@@ -1078,7 +1078,7 @@ void attn_bwd_dv_kernel(TView q, TView k,
   const int wave = threadIdx.x >> 6;
 
   const bf16_t* k_g = k.base(batch, head) + (long)ktile * BK * k.rs;
-  const bf16_t* q_g = q.base(batch, head);
+  const bf16_t* q_g = q.base(batch / q_repeat, head);
   const bf16_t* do_g = dout.base(batch, head);
   const float* lse_g = lse + (long)bh * Lq;
   const bf16_t* bias_g = HAS_BIAS
@@ -1214,7 +1214,7 @@ void attn_bwd_dk_kernel(TView q, TView k, TView v,
                         TViewMut dk,
                         float* __restrict__ dbias,
                         int dbias_chunks, long dbias_stride,
-                        int Lq, int Lk, int heads, int bias_repeat,
+                        int Lq, int Lk, int heads, int bias_repeat, int q_repeat,
                         float scale) {
   __shared__ char k_lds[BK * ROWB];
   __shared__ char v_lds[BK * ROWB];
@@ -1239,7 +1239,7 @@ void attn_bwd_dk_kernel(TView q, TView k, TView v,
 
   const bf16_t* k_g = k.base(batch, head) + (long)ktile * BK * k.rs;
   const bf16_t* v_g = v.base(batch, head) + (long)ktile * BK * v.rs;
-  const bf16_t* q_g = q.base(batch, head);
+  const bf16_t* q_g = q.base(batch / q_repeat, head);
   const bf16_t* do_g = dout.base(batch, head);
   const float* lse_g = lse + (long)bh * Lq;
   const float* delta_g = delta + (long)bh * Lq;
@@ -1439,9 +1439,12 @@ TViewMut make_view_mut(at::Tensor& t) {
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
-                                 long bias_repeat, double scale) {
+                                 long bias_repeat, double scale,
+                                 long q_repeat) {
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
-  const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
+  const int B = k.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
+  TORCH_CHECK(q.size(0) * q_repeat == B,
+              "q batch must be B / q_repeat");
 
   // out in (B, Lq, H, DH) memory (the layout downstream Linears want),
   // exposed as a (B, H, Lq, DH) strided view
@@ -1474,7 +1477,7 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
       has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
       has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
       ov, lse.data_ptr<float>(),                                              \
-      Lq, Lk, H, (int)bias_repeat, (float)scale)
+      Lq, Lk, H, (int)bias_repeat, (int)q_repeat, (float)scale)
 
   if (has_bias && has_mask) DISPATCH(true, true);
   else if (has_bias) DISPATCH(true, false);
@@ -1489,11 +1492,13 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
                                  long bias_repeat, double scale,
-                                 bool need_dbias,
+                                 bool need_dbias, long q_repeat,
                                  c10::optional<at::Tensor> dq_out,
                                  c10::optional<at::Tensor> dk_out,
                                  c10::optional<at::Tensor> dv_out) {
-  const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
+  const int B = k.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
+  TORCH_CHECK(q.size(0) * q_repeat == B,
+              "q batch must be B / q_repeat");
   if (dout.stride(3) != 1) dout = dout.contiguous();
 
   auto delta = at::empty({B, H, Lq}, lse.options());
@@ -1552,7 +1557,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
       has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
       has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
       dov, lse.data_ptr<float>(), delta.data_ptr<float>(), dqv,               \
-      Lq, Lk, H, (int)bias_repeat, (float)scale)
+      Lq, Lk, H, (int)bias_repeat, (int)q_repeat, (float)scale)
 
   if (has_bias && has_mask) DISPATCH_DQ(true, true);
   else if (has_bias) DISPATCH_DQ(true, false);
@@ -1568,7 +1573,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
       dov, lse.data_ptr<float>(), delta.data_ptr<float>(), dkv, dvv,          \
       DB ? dbias.data_ptr<float>() : nullptr,                                 \
       dbias_chunks, need_dbias ? dbias.stride(0) : 0L,                        \
-      Lq, Lk, H, (int)bias_repeat, (float)scale)
+      Lq, Lk, H, (int)bias_repeat, (int)q_repeat, (float)scale)
 
 #define DISPATCH_DV(HB, HM)                                                   \
   hipLaunchKernelGGL((attn_bwd_dv_kernel<HB, HM>), grid_k, dim3(256), 0,      \
@@ -1576,7 +1581,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
       has_bias ? reinterpret_cast<const bf16_t*>(bias_c.data_ptr()) : nullptr,\
       has_mask ? mask_u8.data_ptr<unsigned char>() : nullptr,                 \
       dov, lse.data_ptr<float>(), dvv,                                        \
-      Lq, Lk, H, (int)bias_repeat, (float)scale)
+      Lq, Lk, H, (int)bias_repeat, (int)q_repeat, (float)scale)
 
 #define DISPATCH_DK(HB, HM, DB)                                               \
   hipLaunchKernelGGL((attn_bwd_dk_kernel<HB, HM, DB>), grid_k, dim3(256), 0,  \
@@ -1586,7 +1591,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
       dov, lse.data_ptr<float>(), delta.data_ptr<float>(), dkv,               \
       DB ? dbias.data_ptr<float>() : nullptr,                                 \
       dbias_chunks, need_dbias ? dbias.stride(0) : 0L,                        \
-      Lq, Lk, H, (int)bias_repeat, (float)scale)
+      Lq, Lk, H, (int)bias_repeat, (int)q_repeat, (float)scale)
 
   // split dV/dK passes (4/3 waves per SIMD, but prefetch-free and
   // +40% flops): measured SLOWER end-to-end than the combined kernel

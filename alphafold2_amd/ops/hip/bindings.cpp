@@ -30,13 +30,14 @@ at::Tensor wgrad(at::Tensor dY, at::Tensor X);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
-                                 long bias_repeat, double scale);
+                                 long bias_repeat, double scale,
+                                 long q_repeat);
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  at::Tensor v, at::Tensor out, at::Tensor lse,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
                                  long bias_repeat, double scale,
-                                 bool need_dbias,
+                                 bool need_dbias, long q_repeat,
                                  c10::optional<at::Tensor> dq_out,
                                  c10::optional<at::Tensor> dk_out,
                                  c10::optional<at::Tensor> dv_out);
@@ -68,11 +69,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dY"), py::arg("X"));
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (gfx950)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
-        py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"));
+        py::arg("mask"), py::arg("bias_repeat"), py::arg("scale"),
+        py::arg("q_repeat") = 1);
   m.def("attn_bwd", &attn_bwd, "fused flash attention backward (gfx950)",
         py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
         py::arg("out"), py::arg("lse"), py::arg("bias"), py::arg("mask"),
         py::arg("bias_repeat"), py::arg("scale"), py::arg("need_dbias"),
+        py::arg("q_repeat") = 1,
         py::arg("dq_out") = c10::nullopt, py::arg("dk_out") = c10::nullopt,
         py::arg("dv_out") = c10::nullopt);
 }

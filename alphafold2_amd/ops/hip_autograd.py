@@ -203,14 +203,14 @@ class _AttentionTiedFn(torch.autograd.Function):
         ext = _load_ext()
         Bh = q.shape[0]
         b = Bh // tie_dim
-        qm = q.reshape(b, tie_dim, *q.shape[1:]).mean(dim=1, keepdim=True)
-        q_exp = qm.expand(b, tie_dim, *qm.shape[2:]) \
-                  .reshape(Bh, *qm.shape[2:]).contiguous()
+        # group-mean query, consumed via the kernels' q_repeat divisor —
+        # the tie_dim-fold replication is never materialized
+        qm = q.reshape(b, tie_dim, *q.shape[1:]).mean(dim=1).contiguous()
         bias_c = bias.contiguous() if bias is not None else None
         mask_c = mask.contiguous() if mask is not None else None
-        out, lse = ext.attn_fwd(q_exp, k, v, bias_c, mask_c, bias_repeat,
-                                scale)
-        ctx.save_for_backward(q_exp, k, v, out, lse,
+        out, lse = ext.attn_fwd(qm, k, v, bias_c, mask_c, bias_repeat,
+                                scale, q_repeat=tie_dim)
+        ctx.save_for_backward(qm, k, v, out, lse,
                               *([bias_c] if bias_c is not None else []))
         ctx.has_bias = bias_c is not None
         ctx.mask = mask_c
@@ -222,12 +222,13 @@ class _AttentionTiedFn(torch.autograd.Function):
     def backward(ctx, dout):
         ext = _load_ext()
         saved = ctx.saved_tensors
-        q_exp, k, v, out, lse = saved[:5]
+        qm, k, v, out, lse = saved[:5]
         bias = saved[5] if ctx.has_bias else None
         tie_dim, bias_repeat, scale = ctx.meta
         need_dbias = ctx.bias_requires_grad
-        rets = ext.attn_bwd(dout, q_exp, k, v, out, lse, bias,
-                            ctx.mask, bias_repeat, scale, need_dbias)
+        rets = ext.attn_bwd(dout, qm, k, v, out, lse, bias,
+                            ctx.mask, bias_repeat, scale, need_dbias,
+                            q_repeat=tie_dim)
         dq_exp, dk, dv = rets[:3]
         Bh = dq_exp.shape[0]
         b = Bh // tie_dim
