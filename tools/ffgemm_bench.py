@@ -97,5 +97,20 @@ def main():
               flush=True)
 
 
+    print('--- geglu (fused activation kernels) ---', flush=True)
+    for name, rows, H in [('geglu_fwd_ff1', 5 * 256 * 256, 2048),
+                          ('geglu_bwd_ff1', 5 * 256 * 256, 2048),
+                          ('geglu_bwd_h1024', 5 * 256 * 256, 1024)]:
+        xg = torch.randn(rows, 2 * H, device=dev, dtype=dt)
+        dyg = torch.randn(rows, H, device=dev, dtype=dt)
+        gb = (rows * H * 3 + rows * H) * 2 / 1e9  # ~GB moved
+        if 'fwd' in name:
+            t = time_fn(lambda: ext.geglu_fwd(xg))
+        else:
+            t = time_fn(lambda: ext.geglu_bwd(dyg, xg))
+        print(f'{name:16s} rows={rows:7d} H={H:5d}  {t:.3f}ms '
+              f'(~{gb / t:.1f} TB/s)', flush=True)
+
+
 if __name__ == '__main__':
     main()
