@@ -5,7 +5,6 @@ and utils.py:718-761 (center_distogram_torch).  The cdist+bucketize hot
 path has a fused HIP kernel on MI355X (ops.distance_buckets); this module
 is the always-available eager path and the CPU reference.
 """
-import numpy as np
 import torch
 
 from .. import constants
@@ -33,48 +32,57 @@ def get_bucketed_distance_matrix(coords, mask,
     return buckets
 
 
+def _bin_centers(bins, device):
+    """Representative distance per distogram class.
+
+    Interior classes use the midpoint below their threshold.  The two
+    ends follow the reference's calibration (utils.py:718-761, kept for
+    numerics parity with models trained against it): the first class
+    maps to 1.5 A and the catch-all above-last class to 1.33x the last
+    threshold.
+    """
+    width = bins[2] - bins[1]
+    centers = (bins - 0.5 * width).to(device)
+    centers[0] = 1.5
+    centers[-1] = 1.33 * bins[-1]
+    return centers
+
+
 def center_distogram_torch(distogram, bins=DISTANCE_THRESHOLDS, min_t=1.,
                            center="mean", wide="std"):
     """Distogram (b, N, N, B) -> (central (b,N,N), weights (b,N,N)).
 
-    Central estimate (mean or median over bin midpoints) and a 0-1 weight
-    map derived from the dispersion, with the diagonal and the
-    above-last-threshold class zeroed.  Mirrors reference
-    utils.py:718-761 semantics.
+    Central estimate (probability-weighted mean, or median class) plus a
+    0-1 confidence weight from the dispersion; the diagonal and pairs
+    landing in the catch-all (beyond-range) class get weight 0.
     """
-    shape, device = distogram.shape, distogram.device
-    # bin centers (midpoint below each threshold); clamp the two ends
-    n_bins = (bins - 0.5 * (bins[2] - bins[1])).to(device)
-    n_bins[0] = 1.5
-    n_bins[-1] = 1.33 * bins[-1]  # catch-all class above the last threshold
-    max_bin_allowed = torch.tensor(n_bins.shape[0] - 1, device=device).long()
+    from .backend import as_batched
+    device = distogram.device
+    centers = _bin_centers(bins, device)
+    total = distogram.sum(dim=-1)
 
-    magnitudes = distogram.sum(dim=-1)
     if center == "median":
-        cum_dist = torch.cumsum(distogram, dim=-1)
-        medium = 0.5 * cum_dist[..., -1:]
-        central = torch.searchsorted(cum_dist, medium).squeeze()
-        central = n_bins[torch.min(central, max_bin_allowed)]
-    else:  # mean
-        central = (distogram * n_bins).sum(dim=-1) / magnitudes
-
-    # mask out the catch-all last class
-    mask = (central <= bins[-2].item()).float()
-
-    diag_idxs = np.arange(shape[-2])
-    from .backend import expand_dims_to
-    central = expand_dims_to(central, 3 - len(central.shape))
-    central[:, diag_idxs, diag_idxs] *= 0.
-
-    if wide == "var":
-        dispersion = (distogram * (n_bins - central.unsqueeze(-1)) ** 2).sum(dim=-1) / magnitudes
-    elif wide == "std":
-        dispersion = ((distogram * (n_bins - central.unsqueeze(-1)) ** 2).sum(dim=-1) / magnitudes).sqrt()
+        cdf = torch.cumsum(distogram, dim=-1)
+        median_class = torch.searchsorted(cdf, 0.5 * cdf[..., -1:]) \
+            .squeeze(-1).clamp(max=centers.shape[0] - 1)
+        central = centers[median_class]
     else:
-        dispersion = torch.zeros_like(central, device=device)
+        central = (distogram * centers).sum(dim=-1) / total
 
-    # lower dispersion -> weight closer to 1; nan-safe; zero diagonal
-    weights = mask / (1 + dispersion)
-    weights[weights != weights] *= 0.
-    weights[:, diag_idxs, diag_idxs] *= 0.
+    central = as_batched(central, 3)
+    eye = torch.arange(central.shape[-1], device=device)
+    central[:, eye, eye] = 0.
+
+    in_range = (central <= bins[-2].item()).to(central.dtype)
+    if wide in ("var", "std"):
+        spread = (distogram * (centers - central.unsqueeze(-1)).square()) \
+            .sum(dim=-1) / total
+        if wide == "std":
+            spread = spread.sqrt()
+    else:
+        spread = torch.zeros_like(central)
+
+    weights = in_range / (1 + spread)
+    weights = torch.nan_to_num(weights, nan=0.0)
+    weights[:, eye, eye] = 0.
     return central, weights
