@@ -88,10 +88,13 @@ def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
     and expanded explicitly only on the eager path.  Attention-prob
     dropout (reference alphafold2.py:172) routes to the eager path."""
     drop_active = dropout > 0. and training
+    # head dims < 64 run on the 64-wide tile via zero padding (see
+    # hip_attention_core); only dim_head > 64 or non-multiple-of-8
+    # layouts fall back to eager
     fusable = (
         not drop_active
         and q.dtype == torch.bfloat16
-        and q.shape[-1] == 64
+        and q.shape[-1] <= 64 and q.shape[-1] % 8 == 0
         and (bias is None or bias.dtype == torch.bfloat16)
         and (tie_dim is None or q.shape[0] % tie_dim == 0)
         and using_hip(q, 'attn_fwd')
@@ -103,8 +106,9 @@ def attention_core(q, k, v, bias=None, mask=None, context_mask=None,
                                   tie_dim=tie_dim, bias_repeat=bias_repeat)
     if q.is_cuda and hip_ops_available() and not _FORCE_EAGER:
         reason = ('attention-prob dropout active' if drop_active
-                  else f'dim_head={q.shape[-1]} (fused kernel is 64-wide)'
-                  if q.shape[-1] != 64
+                  else f'dim_head={q.shape[-1]} (fused kernel covers '
+                       'multiples of 8 up to 64)'
+                  if not (q.shape[-1] <= 64 and q.shape[-1] % 8 == 0)
                   else f'dtype={q.dtype} (fused kernel is bf16)')
         warn_gpu_fallback('attention_core', reason)
     if bias is not None and bias_repeat != 1:

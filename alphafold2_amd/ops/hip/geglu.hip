@@ -92,7 +92,7 @@ at::Tensor geglu_fwd(at::Tensor x) {
       row_shift = __builtin_ctz(colt);                                   \
     const int rpi = row_shift ? block >> row_shift : 1;                  \
     long grid = (rows + rpi - 1) / rpi;                                  \
-    if (grid > 8192) grid = 8192;                                        \
+    if (grid > 65536) grid = 65536;                                        \
     if (grid < 1) grid = 1;                                              \
     hipLaunchKernelGGL((geglu_fwd_kernel<T, VEC>), dim3(grid),           \
                        dim3(block), 0, stream,                           \
@@ -141,7 +141,7 @@ at::Tensor geglu_bwd(at::Tensor dy, at::Tensor x) {
       row_shift = __builtin_ctz(colt);                                   \
     const int rpi = row_shift ? block >> row_shift : 1;                  \
     long grid = (rows + rpi - 1) / rpi;                                  \
-    if (grid > 8192) grid = 8192;                                        \
+    if (grid > 65536) grid = 65536;                                        \
     if (grid < 1) grid = 1;                                              \
     hipLaunchKernelGGL((geglu_bwd_kernel<T, VEC>), dim3(grid),           \
                        dim3(block), 0, stream,                           \

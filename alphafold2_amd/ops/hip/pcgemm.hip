@@ -90,19 +90,16 @@ void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
   const int m_rows = min(TM, M - m0);
   const int n_rows = min(TN, N - n0);
 
-  // async-stage (guide T14): the strided 16-byte channel gathers are
-  // the bottleneck (MfmaUtil 3%, VALUBusy 6% — latency-bound), so each
-  // k-tile's loads are issued into registers one iteration EARLY and
-  // only the LDS scatter happens at the barrier; the next tile's
-  // gathers then fly under the MFMA cluster.
-  bf16x8_t a_regs[4], b_regs[4];
-  auto load_tile = [&](int k0) {
+  for (int k0 = 0; k0 < K; k0 += TK) {
+    const int k_rows = min(TK, K - k0);
+    __syncthreads();
+    // stage A and B tiles: 64x32 (row, k) elements each, one 16-byte
+    // 8-channel load per element, scattered into the 8 LDS planes
 #pragma unroll
     for (int pass = 0; pass < 4; ++pass) {
-      const int idx = threadIdx.x + pass * NTHREADS;  // 0..2047
-      const int r = idx >> 5;                         // row in tile
-      const int kk = idx & 31;
-      const int k_rows = min(TK, K - k0);
+      int idx = threadIdx.x + pass * NTHREADS;  // 0..2047
+      int r = idx >> 5;                         // row in tile
+      int kk = idx & 31;
       bf16x8_t av = {};
       bf16x8_t bv = {};
       if (r < m_rows && kk < k_rows) {
@@ -113,27 +110,13 @@ void pcgemm_kernel(const pb16* __restrict__ A, const pb16* __restrict__ B,
         bv = *reinterpret_cast<const bf16x8_t*>(
             Bb_ + (long)(n0 + r) * b_ns + (long)(k0 + kk) * b_ks);
       }
-      a_regs[pass] = av;
-      b_regs[pass] = bv;
-    }
-  };
-
-  load_tile(0);
-  for (int k0 = 0; k0 < K; k0 += TK) {
-    __syncthreads();
-#pragma unroll
-    for (int pass = 0; pass < 4; ++pass) {
-      const int idx = threadIdx.x + pass * NTHREADS;
-      const int r = idx >> 5;
-      const int kk = idx & 31;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        a_lds[j * PLANE + r * PLANE_ROW + kk] = a_regs[pass][j];
-        b_lds[j * PLANE + r * PLANE_ROW + kk] = b_regs[pass][j];
+        a_lds[j * PLANE + r * PLANE_ROW + kk] = av[j];
+        b_lds[j * PLANE + r * PLANE_ROW + kk] = bv[j];
       }
     }
     __syncthreads();
-    if (k0 + TK < K) load_tile(k0 + TK);
 
     // MFMA: each wave owns one d channel of the full 64x64 tile
     const pb16* ap = a_lds + wave * PLANE;

@@ -171,7 +171,7 @@ class _AttentionFn(torch.autograd.Function):
 
 
 def hip_attention_core(q, k, v, bias=None, mask=None, context_mask=None,
-                       tie_dim=None, bias_repeat=1):
+                       tie_dim=None, bias_repeat=1, scale=None):
     key_mask = context_mask if context_mask is not None else mask
     if key_mask is not None:
         key_mask = key_mask.to(torch.uint8)
@@ -179,7 +179,20 @@ def hip_attention_core(q, k, v, bias=None, mask=None, context_mask=None,
             # the kernel indexes mask[batch * Lk + j]: broadcast masks
             # (e.g. the (1, Lk) cross-attention default) expand here
             key_mask = key_mask.expand(q.shape[0], -1).contiguous()
-    scale = q.shape[-1] ** -0.5
+    dh = q.shape[-1]
+    if scale is None:
+        scale = dh ** -0.5
+    if dh < 64:
+        # narrow heads run on the 64-wide MFMA tile via zero padding:
+        # QK^T and the real output channels of PV are invariant to
+        # zero-padded channels, and autograd differentiates the
+        # pad/slice pair (scale already fixed to the REAL head dim)
+        pad = 64 - dh
+        q, k, v = (torch.nn.functional.pad(t, (0, pad)) for t in (q, k, v))
+        out = hip_attention_core(q, k, v, bias=bias, mask=None,
+                                 context_mask=key_mask, tie_dim=tie_dim,
+                                 bias_repeat=bias_repeat, scale=scale)
+        return out[..., :dh]
     if tie_dim is not None:
         return _AttentionTiedFn.apply(q, k, v, bias, key_mask, tie_dim,
                                       bias_repeat, scale)
