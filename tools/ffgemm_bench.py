@@ -112,5 +112,17 @@ def main():
               f'(~{gb / t:.1f} TB/s)', flush=True)
 
 
+    # transcendental-cost probes: same element counts, known traffic
+    xp = torch.randn(327680, 2048, device=dev, dtype=dt)
+    t_gelu = time_fn(lambda: torch.nn.functional.gelu(xp))
+    t_gelut = time_fn(lambda: torch.nn.functional.gelu(xp, approximate='tanh'))
+    t_sig = time_fn(lambda: torch.sigmoid(xp))
+    t_add = time_fn(lambda: xp + xp)
+    gbp = 327680 * 2048 * 2 * 2 / 1e9
+    print(f'probe: torch gelu(erf)={t_gelu:.3f}ms ({gbp/t_gelu:.1f} TB/s)  '
+          f'gelu(tanh)={t_gelut:.3f}ms  sigmoid={t_sig:.3f}ms  '
+          f'add={t_add:.3f}ms', flush=True)
+
+
 if __name__ == '__main__':
     main()
