@@ -278,12 +278,10 @@ class TriangleMultiplicativeModule(nn.Module):
                               self.out_gate.bias])
         hdim = self.left_proj.weight.shape[0]
         fused = F.linear(x, w, bias_cat)
-        left, right, lg, rg, og = fused.split([hdim] * 5, dim=-1)
-
-        # the pair mask folds into the fused gate kernel (no separate
-        # full-tensor mask-multiply passes)
-        left = ops.softclamp_gate(left, lg, row_mask=mask)
-        right = ops.softclamp_gate(right, rg, row_mask=mask)
+        # packed gated projections: the pair mask folds into the gate
+        # kernel, and the backward writes both gatemul gradients into
+        # one packed buffer (no SplitBackward concatenation)
+        left, right, og = ops.tri_proj_gates(fused, hdim, row_mask=mask)
 
         out = ops.triangle_mix(left, right, self.mix)
 

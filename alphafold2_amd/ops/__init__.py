@@ -24,5 +24,6 @@ from .dispatch import (  # noqa: F401
     hip_ops_available, using_hip,
     attention_core, attention_core_packed, geglu, outer_product_mean, triangle_mix,
     pair_outer_sum, distance_buckets, layer_norm, softclamp_gate,
+    tri_proj_gates,
     fused_linear, ff1_geglu,
 )

@@ -229,6 +229,26 @@ def layer_norm(x, weight, bias, eps=1e-5):
     return eager.layer_norm(x, weight, bias, eps)
 
 
+def tri_proj_gates(fused, hdim, row_mask=None):
+    """TriangleMultiplicative's [left|right|lg|rg|og] projection -> the
+    two gated operands + the out-gate slice, as ONE autograd unit whose
+    backward writes into a packed buffer (kills the 5-way SplitBackward
+    concatenation)."""
+    ok = (fused.dtype == torch.bfloat16 and fused.shape[-1] == 5 * hdim
+          and hdim % 8 == 0 and using_hip(fused, 'gatemul_bwd'))
+    if ok:
+        from .hip_autograd import hip_tri_proj_gates
+        return hip_tri_proj_gates(fused, hdim, row_mask)
+    left, right, lg, rg, og = fused.split([hdim] * 5, dim=-1)
+    left = eager.softclamp_gate(left, lg)
+    right = eager.softclamp_gate(right, rg)
+    if row_mask is not None:
+        m = row_mask.unsqueeze(-1).to(left.dtype)
+        left = left * m
+        right = right * m
+    return left, right, og
+
+
 def softclamp_gate(x, gates, row_mask=None):
     """out = x * sigmoid(gates) (* row_mask broadcast per row).
     row_mask: optional bool with shape == x.shape[:-1]."""

@@ -19,7 +19,10 @@ at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs,
                        c10::optional<at::Tensor> rowmask);
 std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x, at::Tensor g,
                                     long xs, long gs,
-                                    c10::optional<at::Tensor> rowmask);
+                                    c10::optional<at::Tensor> rowmask,
+                                    c10::optional<at::Tensor> dx_out,
+                                    c10::optional<at::Tensor> dg_out,
+                                    long dxs, long dgs);
 at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
                       c10::optional<at::Tensor> bias,
                       c10::optional<at::Tensor> resid, long stage);
@@ -55,7 +58,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("rowmask") = c10::nullopt);
   m.def("gatemul_bwd", &gatemul_bwd, "fused x*sigmoid(g) backward (gfx950)",
         py::arg("dy"), py::arg("x"), py::arg("g"), py::arg("xs"),
-        py::arg("gs"), py::arg("rowmask") = c10::nullopt);
+        py::arg("gs"), py::arg("rowmask") = c10::nullopt,
+        py::arg("dx_out") = c10::nullopt, py::arg("dg_out") = c10::nullopt,
+        py::arg("dxs") = 0, py::arg("dgs") = 0);
   m.def("linear_fwd", &linear_fwd,
         "tall-M small-K linear GEMM, bias/residual epilogue (gfx950 MFMA)",
         py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt,

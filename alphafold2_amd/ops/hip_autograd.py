@@ -469,3 +469,65 @@ def hip_outer_product_mean(left, right, mask=None, eps=1e-5):
         count = torch.einsum('b m i, b m j -> b i j', fmask, fmask)
         return outer_sum / (count[..., None] + eps)
     return _OuterSumFn.apply(left.contiguous(), right.contiguous(), 1.0 / m)
+
+
+class _TriProjGatesFn(torch.autograd.Function):
+    """TriangleMultiplicative's gated projections as ONE unit over the
+    fused [left | right | lgate | rgate | ogate] projection (…, 5h):
+
+        gated_left  = left  * sigmoid(lgate) * mask
+        gated_right = right * sigmoid(rgate) * mask
+        og          = ogate                       (passthrough slice)
+
+    The forward reads strided slices (no copies, as before); the win is
+    the BACKWARD: both gatemul gradients are written straight into
+    slices of one packed d_fused buffer, so autograd never runs the
+    5-way SplitBackward concatenation (~16 ms/step at batch 5)."""
+
+    @staticmethod
+    def forward(ctx, fused, hdim, rowmask):
+        ext = _load_ext()
+        C5 = fused.shape[-1]
+        assert C5 == 5 * hdim
+        fused = fused.contiguous()
+        rm = rowmask.reshape(-1).to(torch.uint8).contiguous() \
+            if rowmask is not None else None
+
+        def sl(i):
+            return fused.narrow(-1, i * hdim, hdim)
+
+        left = ext.gatemul_fwd(sl(0), sl(2), C5, C5, rm)
+        right = ext.gatemul_fwd(sl(1), sl(3), C5, C5, rm)
+        ctx.save_for_backward(fused)
+        ctx.hdim = hdim
+        ctx.rm = rm
+        return left, right, sl(4)
+
+    @staticmethod
+    def backward(ctx, d_left, d_right, d_og):
+        ext = _load_ext()
+        (fused,) = ctx.saved_tensors
+        h = ctx.hdim
+        C5 = fused.shape[-1]
+        d_fused = torch.empty_like(fused)
+
+        def sl(t, i):
+            return t.narrow(-1, i * h, h)
+
+        ext.gatemul_bwd(d_left.contiguous(), sl(fused, 0), sl(fused, 2),
+                        C5, C5, ctx.rm,
+                        dx_out=sl(d_fused, 0), dg_out=sl(d_fused, 2),
+                        dxs=C5, dgs=C5)
+        ext.gatemul_bwd(d_right.contiguous(), sl(fused, 1), sl(fused, 3),
+                        C5, C5, ctx.rm,
+                        dx_out=sl(d_fused, 1), dg_out=sl(d_fused, 3),
+                        dxs=C5, dgs=C5)
+        if d_og is not None:
+            sl(d_fused, 4).copy_(d_og)
+        else:
+            sl(d_fused, 4).zero_()
+        return d_fused, None, None
+
+
+def hip_tri_proj_gates(fused, hdim, rowmask=None):
+    return _TriProjGatesFn.apply(fused, hdim, rowmask)
