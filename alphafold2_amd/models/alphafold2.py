@@ -293,9 +293,9 @@ class Alphafold2(nn.Module):
         else:
             raise ValueError('either MSA or embedds must be given')
 
-        # pairwise representation: outer sum + relative position embedding
+        # pairwise representation: outer sum + relative position
+        # embedding, built in ONE fused pass (K13) when on GPU
         x_left, x_right = self.to_pairwise_repr(x).chunk(2, dim=-1)
-        x = ops.pair_outer_sum(x_left, x_right)  # (b, i, j, d)
         x_mask = mask[:, :, None] * mask[:, None, :] if exists(mask) else None
 
         seq_index = default(
@@ -303,7 +303,8 @@ class Alphafold2(nn.Module):
         seq_rel_dist = seq_index[None, :, None] - seq_index[None, None, :]
         seq_rel_dist = seq_rel_dist.clamp(
             -self.max_rel_dist, self.max_rel_dist) + self.max_rel_dist
-        x = x + self.pos_emb(seq_rel_dist)
+        x = ops.pair_rep_build(x_left.contiguous(), x_right.contiguous(),
+                               self.pos_emb.weight, seq_rel_dist)
 
         # recycling inputs
         if exists(recyclables):

@@ -23,7 +23,7 @@ AF2AMD_ALLOW_EAGER_GPU=1 to override.
 from .dispatch import (  # noqa: F401
     hip_ops_available, using_hip,
     attention_core, attention_core_packed, geglu, outer_product_mean, triangle_mix,
-    pair_outer_sum, distance_buckets, layer_norm, softclamp_gate,
+    pair_outer_sum, pair_rep_build, distance_buckets, layer_norm, softclamp_gate,
     tri_proj_gates,
     fused_linear, ff1_geglu,
 )

@@ -215,6 +215,21 @@ def pair_outer_sum(x_left, x_right):
     return eager.pair_outer_sum(x_left, x_right)
 
 
+def pair_rep_build(x_left, x_right, emb_weight, rel):
+    """K13: out[b,i,j] = left[i] + right[j] + emb[rel[i,j]] in one pass
+    (the eager composition materializes the outer sum AND the gathered
+    embedding before adding)."""
+    ok = (x_left.shape[-1] % 8 == 0
+          and _bf16_ok(x_left, x_right, emb_weight)
+          and using_hip(x_left, 'pairrep_fwd'))
+    if ok:
+        from .hip_autograd import hip_pair_rep
+        rel = rel.expand(x_left.shape[0], -1, -1).contiguous().long()
+        return hip_pair_rep(x_left, x_right, emb_weight, rel)
+    return eager.pair_outer_sum(x_left, x_right) \
+        + torch.nn.functional.embedding(rel, emb_weight)
+
+
 def distance_buckets(coords, boundaries):
     if using_hip(coords, 'dist_buckets'):
         ext = _load_ext()

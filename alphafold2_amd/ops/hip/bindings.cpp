@@ -30,6 +30,8 @@ std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
                                       c10::optional<at::Tensor> bias,
                                       long stage);
 at::Tensor wgrad(at::Tensor dY, at::Tensor X);
+at::Tensor pairrep_fwd(at::Tensor left, at::Tensor right, at::Tensor emb,
+                       at::Tensor rel);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  c10::optional<at::Tensor> bias,
                                  c10::optional<at::Tensor> mask,
@@ -69,6 +71,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "linear GEMM with fused GEGLU epilogue (gfx950 MFMA)",
         py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt,
         py::arg("stage") = -1);
+  m.def("pairrep_fwd", &pairrep_fwd,
+        "fused pair-rep build: outer sum + rel-pos embedding gather "
+        "(gfx950, K13)", py::arg("left"), py::arg("right"), py::arg("emb"),
+        py::arg("rel"));
   m.def("wgrad", &wgrad,
         "split-K weight-gradient GEMM dY^T @ X (gfx950 MFMA, fp32 out)",
         py::arg("dY"), py::arg("X"));

@@ -531,3 +531,35 @@ class _TriProjGatesFn(torch.autograd.Function):
 
 def hip_tri_proj_gates(fused, hdim, rowmask=None):
     return _TriProjGatesFn.apply(fused, hdim, rowmask)
+
+
+class _PairRepFn(torch.autograd.Function):
+    """Fused pair-rep build (K13): out[b,i,j] = left[i] + right[j] +
+    emb[rel[i,j]].  Forward writes the (b,n,n,d) tensor ONCE (the eager
+    composition takes three full passes); backward is the standard
+    reduction trio (row/col sums + index_add on the embedding)."""
+
+    @staticmethod
+    @custom_fwd(device_type='cuda', cast_inputs=torch.bfloat16)
+    def forward(ctx, left, right, emb, rel):
+        ext = _load_ext()
+        ctx.save_for_backward(rel)
+        ctx.emb_rows = emb.shape[0]
+        return ext.pairrep_fwd(left.contiguous(), right.contiguous(),
+                               emb.contiguous(), rel.contiguous())
+
+    @staticmethod
+    @custom_bwd(device_type='cuda')
+    def backward(ctx, dy):
+        (rel,) = ctx.saved_tensors
+        d = dy.shape[-1]
+        dleft = dy.sum(dim=2)
+        dright = dy.sum(dim=1)
+        demb = torch.zeros(ctx.emb_rows, d, device=dy.device,
+                           dtype=torch.float32)
+        demb.index_add_(0, rel.reshape(-1), dy.reshape(-1, d).float())
+        return dleft, dright, demb.to(dy.dtype), None
+
+
+def hip_pair_rep(left, right, emb, rel):
+    return _PairRepFn.apply(left, right, emb, rel)

@@ -27,6 +27,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "pcgemm.hip"),
         os.path.join(HIP_DIR, "ffgemm.hip"),
         os.path.join(HIP_DIR, "wgrad.hip"),
+        os.path.join(HIP_DIR, "pairrep.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
