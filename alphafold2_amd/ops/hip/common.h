@@ -64,6 +64,37 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* scratch) {
   return total;
 }
 
+
+// ---- explicit wide vector access ----------------------------------------
+// the compiler does not reliably merge unrolled scalar bf16 accesses
+// (guide §5 common-mistake #2); these force float4/float2 moves.
+
+template <typename T, int VEC>
+__device__ __forceinline__ void vload(const T* __restrict__ p, T* dst) {
+  constexpr int BYTES = sizeof(T) * VEC;
+  if constexpr (BYTES == 16) {
+    *reinterpret_cast<float4*>(dst) = *reinterpret_cast<const float4*>(p);
+  } else if constexpr (BYTES == 8) {
+    *reinterpret_cast<float2*>(dst) = *reinterpret_cast<const float2*>(p);
+  } else {
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) dst[k] = p[k];
+  }
+}
+
+template <typename T, int VEC>
+__device__ __forceinline__ void vstore(T* __restrict__ p, const T* src) {
+  constexpr int BYTES = sizeof(T) * VEC;
+  if constexpr (BYTES == 16) {
+    *reinterpret_cast<float4*>(p) = *reinterpret_cast<const float4*>(src);
+  } else if constexpr (BYTES == 8) {
+    *reinterpret_cast<float2*>(p) = *reinterpret_cast<const float2*>(src);
+  } else {
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) p[k] = src[k];
+  }
+}
+
 // exact gelu (erf form) matching torch.nn.functional.gelu default
 __device__ __forceinline__ float gelu_f(float x) {
   return 0.5f * x * (1.0f + erff(x * 0.70710678118654752440f));

@@ -34,11 +34,14 @@ __global__ void gatemul_fwd_kernel(const T* __restrict__ x, long xs,
     T* yr = y + row * (long)C;
     const float mv = RM ? (float)rowmask[row] : 1.f;
     for (int i = lane * VEC; i < C; i += GROUP * VEC) {
+      T xv[VEC], gv[VEC], yv[VEC];
+      vload<T, VEC>(xr + i, xv);
+      vload<T, VEC>(gr + i, gv);
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) {
-        yr[i + k] = from_f32<T>(to_f32(xr[i + k]) * mv *
-                                sigmoid_f(to_f32(gr[i + k])));
-      }
+      for (int k = 0; k < VEC; ++k)
+        yv[k] = from_f32<T>(to_f32(xv[k]) * mv *
+                            sigmoid_f(to_f32(gv[k])));
+      vstore<T, VEC>(yr + i, yv);
     }
   }
 }
@@ -63,14 +66,19 @@ __global__ void gatemul_bwd_kernel(const T* __restrict__ dy,
     T* dgr = dg + row * dgs;
     const float mv = RM ? (float)rowmask[row] : 1.f;
     for (int i = lane * VEC; i < C; i += GROUP * VEC) {
+      T xv[VEC], gv[VEC], dov[VEC], dxv[VEC], dgv[VEC];
+      vload<T, VEC>(xr + i, xv);
+      vload<T, VEC>(gr + i, gv);
+      vload<T, VEC>(dyr + i, dov);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float go = to_f32(dyr[i + k]) * mv;
-        float xv = to_f32(xr[i + k]);
-        float s = sigmoid_f(to_f32(gr[i + k]));
-        dxr[i + k] = from_f32<T>(go * s);
-        dgr[i + k] = from_f32<T>(go * xv * s * (1.f - s));
+        float go = to_f32(dov[k]) * mv;
+        float s = sigmoid_f(to_f32(gv[k]));
+        dxv[k] = from_f32<T>(go * s);
+        dgv[k] = from_f32<T>(go * to_f32(xv[k]) * s * (1.f - s));
       }
+      vstore<T, VEC>(dxr + i, dxv);
+      vstore<T, VEC>(dgr + i, dgv);
     }
   }
 }
@@ -121,13 +129,15 @@ at::Tensor gatemul_fwd(at::Tensor x, at::Tensor g, long xs, long gs,
     else LAUNCH_G(T, VEC, 64);                                              \
   } while (0)
 
+  const bool al8 = (C % 8) == 0 && (xs % 8) == 0 && (gs % 8) == 0;
+  const bool al4 = (C % 4) == 0 && (xs % 4) == 0 && (gs % 4) == 0;
   if (x.scalar_type() == at::kBFloat16) {
-    if ((C % 8) == 0) LAUNCH(__hip_bfloat16, 8);
+    if (al8) LAUNCH(__hip_bfloat16, 8);
     else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if ((C % 4) == 0) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if (al4) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
-    if ((C % 8) == 0) LAUNCH(__half, 8); else LAUNCH(__half, 1);
+    if (al8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "gatemul_fwd: unsupported dtype");
   }
@@ -185,13 +195,17 @@ std::vector<at::Tensor> gatemul_bwd(at::Tensor dy, at::Tensor x,
     else LAUNCH_G(T, VEC, 64);                                              \
   } while (0)
 
+  const bool al8 = (C % 8) == 0 && (xs % 8) == 0 && (gs % 8) == 0
+      && (dxs % 8) == 0 && (dgs % 8) == 0;
+  const bool al4 = (C % 4) == 0 && (xs % 4) == 0 && (gs % 4) == 0
+      && (dxs % 4) == 0 && (dgs % 4) == 0;
   if (x.scalar_type() == at::kBFloat16) {
-    if ((C % 8) == 0) LAUNCH(__hip_bfloat16, 8);
+    if (al8) LAUNCH(__hip_bfloat16, 8);
     else LAUNCH(__hip_bfloat16, 1);
   } else if (x.scalar_type() == at::kFloat) {
-    if ((C % 4) == 0) LAUNCH(float, 4); else LAUNCH(float, 1);
+    if (al4) LAUNCH(float, 4); else LAUNCH(float, 1);
   } else if (x.scalar_type() == at::kHalf) {
-    if ((C % 8) == 0) LAUNCH(__half, 8); else LAUNCH(__half, 1);
+    if (al8) LAUNCH(__half, 8); else LAUNCH(__half, 1);
   } else {
     TORCH_CHECK(false, "gatemul_bwd: unsupported dtype");
   }

@@ -12,35 +12,6 @@
 
 namespace {
 
-// explicit wide vector access: the compiler does not reliably merge
-// unrolled scalar bf16 loads/stores (guide §5 common-mistake #2 — the
-// scalar form measured 1.3-2.1 TB/s vs ~4.3 TB/s for torch's gelu)
-template <typename T, int VEC>
-__device__ __forceinline__ void vload(const T* __restrict__ p, T* dst) {
-  constexpr int BYTES = sizeof(T) * VEC;
-  if constexpr (BYTES == 16) {
-    *reinterpret_cast<float4*>(dst) = *reinterpret_cast<const float4*>(p);
-  } else if constexpr (BYTES == 8) {
-    *reinterpret_cast<float2*>(dst) = *reinterpret_cast<const float2*>(p);
-  } else {
-#pragma unroll
-    for (int k = 0; k < VEC; ++k) dst[k] = p[k];
-  }
-}
-
-template <typename T, int VEC>
-__device__ __forceinline__ void vstore(T* __restrict__ p, const T* src) {
-  constexpr int BYTES = sizeof(T) * VEC;
-  if constexpr (BYTES == 16) {
-    *reinterpret_cast<float4*>(p) = *reinterpret_cast<const float4*>(src);
-  } else if constexpr (BYTES == 8) {
-    *reinterpret_cast<float2*>(p) = *reinterpret_cast<const float2*>(src);
-  } else {
-#pragma unroll
-    for (int k = 0; k < VEC; ++k) p[k] = src[k];
-  }
-}
-
 // row-outer loop with multi-row packing: when H/VEC < blockDim the
 // block covers several rows per iteration (shift-based row split for
 // power-of-2 H — a runtime `idx / H` in the hot loop serializes on the

@@ -43,9 +43,11 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
 
     float sum = 0.f, sumsq = 0.f;
     for (int i = lane * VEC; i < D; i += GROUP * VEC) {
+      T xv[VEC];
+      vload<T, VEC>(xr + i, xv);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float v = to_f32(xr[i + k]);
+        float v = to_f32(xv[k]);
         sum += v;
         sumsq += v * v;
       }
@@ -62,11 +64,14 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
     }
 
     for (int i = lane * VEC; i < D; i += GROUP * VEC) {
+      T xv[VEC], yv[VEC];
+      vload<T, VEC>(xr + i, xv);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float xhat = (to_f32(xr[i + k]) - mean) * rstd;
-        yr[i + k] = from_f32<T>(xhat * w[i + k] + b[i + k]);
+        float xhat = (to_f32(xv[k]) - mean) * rstd;
+        yv[k] = from_f32<T>(xhat * w[i + k] + b[i + k]);
       }
+      vstore<T, VEC>(yr + i, yv);
     }
   }
 }
@@ -107,10 +112,13 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
     for (int chunk = 0; chunk < MAXCHUNK; ++chunk) {
       const int i = (chunk * GROUP + lane) * VEC;
       if (i < D) {
+        T gv[VEC], xv[VEC];
+        vload<T, VEC>(dyr + i, gv);
+        vload<T, VEC>(xr + i, xv);
 #pragma unroll
         for (int k = 0; k < VEC; ++k) {
-          float g = to_f32(dyr[i + k]);
-          float xhat = (to_f32(xr[i + k]) - m) * rs;
+          float g = to_f32(gv[k]);
+          float xhat = (to_f32(xv[k]) - m) * rs;
           float gw = g * w[i + k];
           c1 += gw;
           c2 += gw * xhat;
@@ -136,12 +144,16 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
     c2 = group_sum<GROUP>(c2) / D;
 
     for (int i = lane * VEC; i < D; i += GROUP * VEC) {
+      T gv[VEC], xv[VEC], ov[VEC];
+      vload<T, VEC>(dyr + i, gv);
+      vload<T, VEC>(xr + i, xv);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float g = to_f32(dyr[i + k]);
-        float xhat = (to_f32(xr[i + k]) - m) * rs;
-        dxr[i + k] = from_f32<T>(rs * (g * w[i + k] - c1 - xhat * c2));
+        float g = to_f32(gv[k]);
+        float xhat = (to_f32(xv[k]) - m) * rs;
+        ov[k] = from_f32<T>(rs * (g * w[i + k] - c1 - xhat * c2));
       }
+      vstore<T, VEC>(dxr + i, ov);
     }
   }
 

@@ -43,14 +43,15 @@ __global__ void pairrep_fwd_kernel(const T* __restrict__ left,
   const int jstep = blockDim.x / chunks;
   for (int j = j0; j < n; j += jstep) {
     const long e = relrow[j];
-    const T* rr = rb + (long)j * d + c8;
-    const T* er = emb + e * (long)d + c8;
-    T* orow = og + (long)j * d + c8;
+    T rv[8], ev[8], ov[8];
+    vload<T, 8>(rb + (long)j * d + c8, rv);
+    vload<T, 8>(emb + e * (long)d + c8, ev);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      orow[k] = from_f32<T>(to_f32(lrow[c8 + k]) + to_f32(rr[k]) +
-                            to_f32(er[k]));
+      ov[k] = from_f32<T>(to_f32(lrow[c8 + k]) + to_f32(rv[k]) +
+                          to_f32(ev[k]));
     }
+    vstore<T, 8>(og + (long)j * d + c8, ov);
   }
 }
 
