@@ -65,6 +65,17 @@ def kabsch_numpy(X, Y):
 # losses / metrics
 
 
+def _as_distmat(coords, mat, p, clamp):
+    """Coordinates (N, d) -> pairwise distance matrix, unless a matrix
+    was supplied directly."""
+    if mat is not None:
+        return mat
+    pts = coords.squeeze()
+    if clamp is not None:
+        pts = pts.clamp(*clamp)
+    return torch.cdist(pts, pts, p=p)
+
+
 def distmat_loss_torch(X=None, Y=None, X_mat=None, Y_mat=None, p=2, q=2,
                        custom=None, distmat_mask=None, clamp=None):
     """Distance-matrix loss between predicted and true structures.
@@ -75,25 +86,16 @@ def distmat_loss_torch(X=None, Y=None, X_mat=None, Y_mat=None, p=2, q=2,
     assert (X is not None or X_mat is not None) and \
            (Y is not None or Y_mat is not None), \
         "true and predicted coords or dist mats must be provided"
-    if X_mat is None:
-        X = X.squeeze()
-        if clamp is not None:
-            X = torch.clamp(X, *clamp)
-        X_mat = torch.cdist(X, X, p=p)
-    if Y_mat is None:
-        Y = Y.squeeze()
-        if clamp is not None:
-            Y = torch.clamp(Y, *clamp)
-        Y_mat = torch.cdist(Y, Y, p=p)
-    if distmat_mask is None:
-        distmat_mask = torch.ones_like(Y_mat).bool()
-
+    pred = _as_distmat(X, X_mat, p, clamp)
+    true = _as_distmat(Y, Y_mat, p, clamp)
     if custom is not None:
-        return custom(X_mat.squeeze(), Y_mat.squeeze()).mean()
-    loss = (X_mat - Y_mat) ** 2
+        return custom(pred.squeeze(), true.squeeze()).mean()
+    per_pair = (pred - true).square()
     if q != 2:
-        loss = loss ** (q / 2)
-    return loss[distmat_mask].mean()
+        per_pair = per_pair.pow(q / 2)
+    if distmat_mask is not None:
+        per_pair = per_pair[distmat_mask]
+    return per_pair.mean()
 
 
 def rmsd_torch(X, Y):
