@@ -92,7 +92,7 @@ int gm_group(int C, int VEC) {
 
 long gm_grid(long rows, int rpb) {
   long blocks = (rows + rpb - 1) / rpb;
-  const long cap = 4096;
+  const long cap = 65536;
   return blocks < cap ? blocks : (cap > 0 ? cap : 1);
 }
 
