@@ -90,9 +90,13 @@ def main():
         x = torch.randn(K, N, device=dev, dtype=dt) * 0.1
         fl = 2.0 * K * M * N
         t_blas = time_fn(lambda: dy.t() @ x)
+        t_swap = time_fn(lambda: (x.t() @ dy).t().contiguous())
+        t_swapv = time_fn(lambda: x.t() @ dy)   # consumer-transposed view
         t_mine = time_fn(lambda: ext.wgrad(dy, x))
         print(f'{name:16s} K={K:7d} M={M:4d} N={N:4d}  '
               f'blas={t_blas:.3f}ms ({fl / t_blas / 1e9:.0f} TF)  '
+              f'swap={t_swap:.3f}ms  swapv={t_swapv:.3f}ms '
+              f'({fl / t_swapv / 1e9:.0f} TF)  '
               f'mine={t_mine:.3f}ms ({fl / t_mine / 1e9:.0f} TF)',
               flush=True)
 
