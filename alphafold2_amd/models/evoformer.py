@@ -141,7 +141,8 @@ class Attention(nn.Module):
                     out = out.transpose(-2, -3).reshape(*x.shape[:-1], -1)
                     gates = fused.narrow(-1, 3 * inner, inner)
                     out = ops.softclamp_gate(out, gates)
-                    return self.to_out(out)
+                    return ops.fused_linear(out, self.to_out.weight,
+                                            self.to_out.bias)
 
             q, k, v, gates = fused.split([inner, inner, inner, inner], dim=-1)
         else:
@@ -169,7 +170,7 @@ class Attention(nn.Module):
         if gates is None:
             gates = self.gating(x)
         out = ops.softclamp_gate(out, gates)
-        return self.to_out(out)
+        return ops.fused_linear(out, self.to_out.weight, self.to_out.bias)
 
 
 class AxialAttention(nn.Module):
@@ -287,7 +288,7 @@ class TriangleMultiplicativeModule(nn.Module):
 
         out = self.to_out_norm(out)
         out = ops.softclamp_gate(out, og)
-        return self.to_out(out)
+        return ops.fused_linear(out, self.to_out.weight, self.to_out.bias)
 
 
 # ---------------------------------------------------------------------------
@@ -311,7 +312,8 @@ class OuterMean(nn.Module):
         hdim = self.left_proj.weight.shape[0]
         left, right = F.linear(x, w, bias_cat).split([hdim, hdim], dim=-1)
         outer = ops.outer_product_mean(left, right, mask=mask, eps=self.eps)
-        return self.proj_out(outer)
+        return ops.fused_linear(outer, self.proj_out.weight,
+                                self.proj_out.bias)
 
 
 # ---------------------------------------------------------------------------

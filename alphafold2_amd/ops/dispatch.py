@@ -161,19 +161,16 @@ def _bf16_ok(*tensors):
     return True
 
 
-_CUSTOM_LINEAR = os.environ.get("AF2AMD_CUSTOM_LINEAR", "0") == "1"
-
-
 def fused_linear(x, weight, bias=None, residual=None):
     """out = x @ weight.T (+ bias) (+ residual).
 
     Measured (profiles/r02_ffgemm_ab.log): hipBLASLt beats the custom
-    128x128 GEMM on plain/residual linears (the residual add is cheaper
-    than the GEMM gap), so those stay on Tensile; the custom kernel is
-    dispatch-gated behind AF2AMD_CUSTOM_LINEAR=1 for A/B work.  The
-    GEGLU fusion (ff1_geglu) is the variant that wins."""
-    ok = (_CUSTOM_LINEAR
-          and x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
+    128x128 GEMM on plain/residual FORWARDS, so _LinearFn runs Tensile
+    forward by default (AF2AMD_CUSTOM_LINEAR=1 forces the custom GEMM
+    for A/B work); the routing still pays in backward, where the
+    split-K wgrad kernel takes the small-output/huge-K gradient class
+    Tensile runs at ~98 TF."""
+    ok = (x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
           and _bf16_ok(x, weight, residual)
           and using_hip(x, 'linear_fwd'))
     if ok:

@@ -52,23 +52,52 @@ def _dgrad(dy2, weight):
     return dy2 @ weight
 
 
+def _wgrad(dy2, x2):
+    """dW = dY^T @ X.  The custom split-K kernel wins ONLY the
+    small-output / huge-K class (measured: (256, 512) 0.64 vs 0.88 ms;
+    (2048, 256) and (256, 1024) lose — profiles/r02_ffgemm_ab.log),
+    so dispatch is gated to that class."""
+    M, N = dy2.shape[-1], x2.shape[-1]
+    K = dy2.shape[0]
+    if (M * N <= 256 * 512 and K >= 65536 and M % 8 == 0 and N % 8 == 0
+            and dy2.dtype == torch.bfloat16):
+        ext = _load_ext()
+        return ext.wgrad(dy2.contiguous(), x2.contiguous())
+    return dy2.t() @ x2
+
+
+import os as _os
+
+_FORCE_CUSTOM_LINEAR_FWD = _os.environ.get(
+    "AF2AMD_CUSTOM_LINEAR", "0") == "1"
+
+
 class _LinearFn(torch.autograd.Function):
-    """out = x @ W.T (+ bias) (+ residual) — forward on the fused MFMA
-    GEMM; dgrad reuses the same kernel, wgrad stays on hipBLASLt (its
-    K = M reduction is a different shape class)."""
+    """out = x @ W.T (+ bias) (+ residual).
+
+    Forward runs hipBLASLt (measured faster than the custom GEMM on
+    plain/residual shapes) unless AF2AMD_CUSTOM_LINEAR=1; the value of
+    routing plain Linears through this Function is the BACKWARD: the
+    split-K wgrad kernel takes the small-output/huge-K gradients that
+    Tensile handles poorly (~98 TF on the attention out-projection)."""
 
     @staticmethod
     @custom_fwd(device_type='cuda', cast_inputs=torch.bfloat16)
     def forward(ctx, x, weight, bias, residual):
-        ext = _load_ext()
         x = x.contiguous()
         ctx.save_for_backward(x, weight)
         ctx.has_bias = bias is not None
         ctx.has_resid = residual is not None
-        return ext.linear_fwd(
-            x, weight.contiguous(),
-            bias.contiguous() if bias is not None else None,
-            residual.contiguous() if residual is not None else None)
+        if _FORCE_CUSTOM_LINEAR_FWD:
+            ext = _load_ext()
+            return ext.linear_fwd(
+                x, weight.contiguous(),
+                bias.contiguous() if bias is not None else None,
+                residual.contiguous() if residual is not None else None)
+        out = torch.nn.functional.linear(x, weight, bias)
+        if residual is not None:
+            out = out + residual
+        return out
 
     @staticmethod
     @custom_bwd(device_type='cuda')
@@ -80,7 +109,7 @@ class _LinearFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             dx = _dgrad(dy2, weight).view_as(x)
         if ctx.needs_input_grad[1]:
-            dw = dy2.t() @ x2
+            dw = _wgrad(dy2, x2)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy2.sum(dim=0)
         dr = dy if ctx.has_resid else None
@@ -121,7 +150,7 @@ class _FF1GegluFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             dx = _dgrad(di2, weight).view_as(x)
         if ctx.needs_input_grad[1]:
-            dw = di2.t() @ x2
+            dw = _wgrad(di2, x2)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = di2.sum(dim=0)
         return dx, dw, db
