@@ -85,6 +85,8 @@ def main():
         ('ff2_wgrad_pair', 5 * 256 * 256, 256, 1024),
         ('qkvg_wgrad', 5 * 256 * 256, 2048, 256),
         ('out_wgrad', 5 * 256 * 256, 256, 512),
+        ('trimul_out_wgrad', 5 * 256 * 256, 256, 256),
+        ('gating_wgrad', 5 * 256 * 256, 512, 256),
     ]:
         dy = torch.randn(K, M, device=dev, dtype=dt) * 0.1
         x = torch.randn(K, N, device=dev, dtype=dt) * 0.1
