@@ -50,6 +50,8 @@ def main():
     from alphafold2_amd.utils import get_bucketed_distance_matrix, kabsch_torch
 
     rank, world_size, local_rank = init_distributed()
+    from alphafold2_amd.runtime import enable_tuned_gemm
+    enable_tuned_gemm()
     device = torch.device('cuda', local_rank) if torch.cuda.is_available() \
         else torch.device('cpu')
     if device.type == 'cuda':
