@@ -124,48 +124,6 @@ __device__ __forceinline__ void stage_tile_rowstride(
   }
 }
 
-// async-stage split of the rowstride (bias) staging: the dkv kernel's
-// bias tile otherwise loads inside the barrier window, putting an L2
-// round-trip on the critical path of every q-tile iteration
-struct BiasRegs {
-  float4 v0, v1;
-};
-
-__device__ __forceinline__ bool bias_stage_fast(const bf16_t* g,
-                                                long row_stride) {
-  return (row_stride % 8) == 0 && (((uintptr_t)g) & 15) == 0;
-}
-
-__device__ __forceinline__ void bias_stage_load(
-    const bf16_t* __restrict__ g, long row_stride, int rows, int cols,
-    BiasRegs& r) {
-  const int tid = threadIdx.x;
-#pragma unroll
-  for (int pass = 0; pass < 2; ++pass) {
-    int idx = tid + pass * 256;
-    int row = idx >> 3;
-    int c16 = (idx & 7) << 4;
-    float4 val = {0, 0, 0, 0};
-    if (row < rows && c16 < cols * (int)sizeof(bf16_t)) {
-      val = *reinterpret_cast<const float4*>(
-          reinterpret_cast<const char*>(g + row * row_stride) + c16);
-    }
-    (pass ? r.v1 : r.v0) = val;
-  }
-}
-
-__device__ __forceinline__ void bias_stage_store(const BiasRegs& r,
-                                                 char* lds) {
-  const int tid = threadIdx.x;
-#pragma unroll
-  for (int pass = 0; pass < 2; ++pass) {
-    int idx = tid + pass * 256;
-    int row = idx >> 3;
-    int c16 = (idx & 7) << 4;
-    *reinterpret_cast<float4*>(lds + swz(row, c16)) = pass ? r.v1 : r.v0;
-  }
-}
-
 // stage a [rows<=64][64] bf16 tile TRANSPOSED into LDS ([col][row],
 // swizzled): the PV B-operand wants V^T rows so its fragments become
 // single 16-byte ds_reads instead of 8 scalar strided reads.
@@ -906,15 +864,9 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
   }
 
   const int n_q = (Lq + BQ - 1) / BQ;
-  const int b_cols = min(BK, Lk - ktile * BK);
   StageRegs qreg, doreg;
-  BiasRegs breg;
-  const bool bias_pre = HAS_BIAS && bias_stage_fast(bias_g, Lk);
   stage_load(q_g, q.rs, min(BQ, Lq), qreg);
   stage_load(do_g, dout.rs, min(BQ, Lq), doreg);
-  if (bias_pre)
-    bias_stage_load(bias_g + (long)ktile * BK, Lk, min(BQ, Lq), b_cols,
-                    breg);
   for (int t = 0; t < n_q; ++t) {
     const int q_rows = min(BQ, Lq - t * BQ);
     __syncthreads();
@@ -923,12 +875,8 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
     stage_store(doreg, do_lds);
     stage_store_t(doreg, dot_lds);
     if (HAS_BIAS) {
-      if (bias_pre) {
-        bias_stage_store(breg, b_lds);
-      } else {
-        stage_tile_rowstride(bias_g + (long)t * BQ * Lk + (long)ktile * BK,
-                             Lk, q_rows, b_cols, b_lds);
-      }
+      stage_tile_rowstride(bias_g + (long)t * BQ * Lk + (long)ktile * BK,
+                           Lk, q_rows, min(BK, Lk - ktile * BK), b_lds);
     }
     if (threadIdx.x < BQ) {
       const int qq = t * BQ + threadIdx.x;
@@ -941,9 +889,6 @@ void attn_bwd_dkv_kernel(TView q, TView k, TView v,
       stage_load(q_g + (long)(t + 1) * BQ * q.rs, q.rs, next_rows, qreg);
       stage_load(do_g + (long)(t + 1) * BQ * dout.rs, dout.rs, next_rows,
                  doreg);
-      if (bias_pre)
-        bias_stage_load(bias_g + (long)(t + 1) * BQ * Lk
-                        + (long)ktile * BK, Lk, next_rows, b_cols, breg);
     }
 
     // S^T = K Q^T : A = K (k = d), B col = q row of Q (k = d contiguous)
