@@ -440,3 +440,15 @@ def test_train_pre_scn_data(tmp_path):
         cwd=root, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-2000:]
     assert 'loss' in r.stdout
+
+
+def test_enable_tuned_gemm_cpu_noop():
+    from alphafold2_amd.runtime import enable_tuned_gemm
+    if torch.cuda.is_available():
+        pytest.skip('GPU present')
+    assert enable_tuned_gemm() is False  # no device -> no-op
+    # the shipped cache exists and carries this image's validators
+    import alphafold2_amd.runtime.tuning as tuning
+    assert os.path.exists(tuning._CACHE)
+    head = open(tuning._CACHE).read(200)
+    assert 'Validator' in head and 'GCN_ARCH_NAME' in head
