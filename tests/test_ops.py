@@ -169,3 +169,58 @@ def test_attention_dropout_eager():
         o3 = m(x)
         o4 = m(x)
     assert torch.allclose(o3, o4), 'eval must be deterministic'
+
+
+def test_pair_rep_build_eager_matches_composition():
+    import torch
+    from alphafold2_amd import ops
+    torch.manual_seed(0)
+    b, n, d, V = 2, 12, 16, 9
+    left = torch.randn(b, n, d)
+    right = torch.randn(b, n, d)
+    emb = torch.randn(V, d)
+    rel = torch.randint(0, V, (1, n, n))
+    out = ops.pair_rep_build(left, right, emb, rel)
+    ref = left[:, :, None, :] + right[:, None, :, :] \
+        + torch.nn.functional.embedding(rel, emb)
+    assert torch.allclose(out, ref, atol=1e-6)
+
+
+def test_tri_proj_gates_eager_path():
+    import torch
+    from alphafold2_amd import ops
+    torch.manual_seed(1)
+    b, n, h = 1, 8, 16
+    fused = torch.randn(b, n, n, 5 * h)
+    mask = torch.rand(b, n, n) > 0.3
+    left, right, og = ops.tri_proj_gates(fused, h, row_mask=mask)
+    l0, r0, lg, rg, og0 = fused.split([h] * 5, dim=-1)
+    m = mask.unsqueeze(-1).float()
+    assert torch.allclose(left, l0 * torch.sigmoid(lg) * m, atol=1e-6)
+    assert torch.allclose(right, r0 * torch.sigmoid(rg) * m, atol=1e-6)
+    assert torch.equal(og, og0)
+
+
+def test_fused_linear_eager_residual():
+    import torch
+    from alphafold2_amd import ops
+    torch.manual_seed(2)
+    x = torch.randn(6, 16)
+    w = torch.randn(8, 16)
+    b = torch.randn(8)
+    r = torch.randn(6, 8)
+    out = ops.fused_linear(x, w, b, residual=r)
+    assert torch.allclose(out, x @ w.t() + b + r, atol=1e-5)
+
+
+def test_ff1_geglu_eager_matches_reference_math():
+    import torch
+    from alphafold2_amd import ops
+    torch.manual_seed(3)
+    x = torch.randn(5, 16)
+    w = torch.randn(64, 16)
+    b = torch.randn(64)
+    out = ops.ff1_geglu(x, w, b)
+    i = x @ w.t() + b
+    a, g = i.chunk(2, dim=-1)
+    assert torch.allclose(out, a * torch.nn.functional.gelu(g), atol=1e-5)
