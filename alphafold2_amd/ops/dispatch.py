@@ -216,7 +216,7 @@ def pair_rep_build(x_left, x_right, emb_weight, rel):
     """K13: out[b,i,j] = left[i] + right[j] + emb[rel[i,j]] in one pass
     (the eager composition materializes the outer sum AND the gathered
     embedding before adding)."""
-    ok = (x_left.shape[-1] % 8 == 0
+    ok = (x_left.shape[-1] % 8 == 0 and x_left.shape[-1] <= 2048
           and _bf16_ok(x_left, x_right, emb_weight)
           and using_hip(x_left, 'pairrep_fwd'))
     if ok:

@@ -36,7 +36,9 @@ __global__ void pairrep_fwd_kernel(const T* __restrict__ left,
   const T* rb = right + b * (long)n * d;
   const long* relrow = rel + bi * (long)n;
   T* og = out + bi * (long)n * d;
-  // thread t owns 8-wide chunk (t % chunks) of row j = t / chunks
+  // thread t owns 8-wide chunk (t % chunks) of row j = t / chunks.
+  // When chunks does not divide blockDim the trailing threads duplicate
+  // a row (identical values written twice — benign).
   const int chunks = d / 8;
   const int c8 = (threadIdx.x % chunks) * 8;
   const int j0 = threadIdx.x / chunks;
@@ -66,6 +68,8 @@ at::Tensor pairrep_fwd(at::Tensor left, at::Tensor right, at::Tensor emb,
   TORCH_CHECK(rel.scalar_type() == at::kLong, "pairrep_fwd: rel must be long");
   const int b = left.size(0), n = left.size(1), d = left.size(2);
   TORCH_CHECK(d % 8 == 0, "pairrep_fwd: d must be a multiple of 8");
+  TORCH_CHECK(d <= 2048, "pairrep_fwd: d > 2048 not covered (one block "
+              "row owns d/8 chunks across 256 threads)");
   TORCH_CHECK(emb.size(1) == d && right.size(2) == d, "pairrep_fwd: dim");
   auto out = at::empty({b, n, n, d}, left.options());
   const int block = 256;

@@ -543,14 +543,20 @@ class _TriProjGatesFn(torch.autograd.Function):
         def sl(t, i):
             return t.narrow(-1, i * h, h)
 
-        ext.gatemul_bwd(d_left.contiguous(), sl(fused, 0), sl(fused, 2),
-                        C5, C5, ctx.rm,
-                        dx_out=sl(d_fused, 0), dg_out=sl(d_fused, 2),
-                        dxs=C5, dgs=C5)
-        ext.gatemul_bwd(d_right.contiguous(), sl(fused, 1), sl(fused, 3),
-                        C5, C5, ctx.rm,
-                        dx_out=sl(d_fused, 1), dg_out=sl(d_fused, 3),
-                        dxs=C5, dgs=C5)
+        def bwd_pair(dy, xi, gi):
+            # an output can be unused downstream (None grad): its
+            # operand/gate slices then get zero gradient
+            if dy is None:
+                sl(d_fused, xi).zero_()
+                sl(d_fused, gi).zero_()
+                return
+            ext.gatemul_bwd(dy.contiguous(), sl(fused, xi), sl(fused, gi),
+                            C5, C5, ctx.rm,
+                            dx_out=sl(d_fused, xi), dg_out=sl(d_fused, gi),
+                            dxs=C5, dgs=C5)
+
+        bwd_pair(d_left, 0, 2)
+        bwd_pair(d_right, 1, 3)
         if d_og is not None:
             sl(d_fused, 4).copy_(d_og)
         else:
