@@ -189,6 +189,13 @@ def ff1_geglu(x, weight, bias=None):
           and _bf16_ok(x, weight)
           and using_hip(x, 'ff1_geglu_fwd'))
     if ok:
+        if not torch.is_grad_enabled():
+            # inference: skip materializing the 2x-width pre-activation
+            ext = _load_ext()
+            x16, w16, b16 = (t.to(torch.bfloat16).contiguous()
+                             if t is not None else None
+                             for t in (x, weight, bias))
+            return ext.ff1_geglu_fwd(x16, w16, b16, -1, False)[0]
         from .hip_autograd import hip_ff1_geglu
         return hip_ff1_geglu(x, weight, bias)
     return geglu(torch.nn.functional.linear(x, weight, bias))

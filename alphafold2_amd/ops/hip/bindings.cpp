@@ -28,7 +28,7 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
                       c10::optional<at::Tensor> resid, long stage);
 std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
                                       c10::optional<at::Tensor> bias,
-                                      long stage);
+                                      long stage, bool want_inter);
 at::Tensor wgrad(at::Tensor dY, at::Tensor X);
 at::Tensor pairrep_fwd(at::Tensor left, at::Tensor right, at::Tensor emb,
                        at::Tensor rel);
@@ -70,7 +70,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ff1_geglu_fwd", &ff1_geglu_fwd,
         "linear GEMM with fused GEGLU epilogue (gfx950 MFMA)",
         py::arg("x"), py::arg("W"), py::arg("bias") = c10::nullopt,
-        py::arg("stage") = -1);
+        py::arg("stage") = -1, py::arg("want_inter") = true);
   m.def("pairrep_fwd", &pairrep_fwd,
         "fused pair-rep build: outer sum + rel-pos embedding gather "
         "(gfx950, K13)", py::arg("left"), py::arg("right"), py::arg("emb"),

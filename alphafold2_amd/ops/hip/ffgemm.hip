@@ -211,8 +211,10 @@ void linear_gemm_kernel(const pb16* __restrict__ A,
   __syncthreads();   // staging pool -> C restage
 
   if (EPI == EPI_GEGLU) {
-    // raw pre-activation (val | gate) goes straight to `inter` (needed
-    // by backward); the gated product is restaged for coalesced writes
+    // raw pre-activation (val | gate) goes straight to `inter` when the
+    // caller needs it for backward (inference passes nullptr and skips
+    // the 2x-width write); the gated product is restaged for coalesced
+    // writes
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
@@ -223,7 +225,7 @@ void linear_gemm_kernel(const pb16* __restrict__ A,
           const int col = nt * 16 + fcol;
           const float val = acc[mt][nt][reg] + bval[nt];
           const float gat = acc[mt][nt + 4][reg] + bval[nt + 4];
-          if (row < m_rows && col < n_cols) {
+          if (inter != nullptr && row < m_rows && col < n_cols) {
             inter[(long)(m0 + row) * N + n0 + col] = (pb16)val;
             inter[(long)(m0 + row) * N + half + n0 + col] = (pb16)gat;
           }
@@ -335,7 +337,7 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor W,
 // where inter = x @ W^T + bias is the raw pre-activation (for backward).
 std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
                                       c10::optional<at::Tensor> bias,
-                                      long stage) {
+                                      long stage, bool want_inter) {
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
               W.scalar_type() == at::kBFloat16, "ff1_geglu_fwd: bf16 only");
   TORCH_CHECK(x.is_contiguous() && W.is_contiguous(),
@@ -349,9 +351,12 @@ std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
   auto osz = x.sizes().vec();
   osz.back() = N / 2;
   auto out = at::empty(osz, x.options());
-  auto isz = x.sizes().vec();
-  isz.back() = N;
-  auto inter = at::empty(isz, x.options());
+  at::Tensor inter;
+  if (want_inter) {
+    auto isz = x.sizes().vec();
+    isz.back() = N;
+    inter = at::empty(isz, x.options());
+  }
 
   const int mtiles = (M + BM - 1) / BM;
   const int ntiles = (N / 2 + 63) / 64;
@@ -367,7 +372,8 @@ std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
                      reinterpret_cast<const pb16*>(x.data_ptr()),        \
                      reinterpret_cast<const pb16*>(W.data_ptr()), bp,    \
                      nullptr, reinterpret_cast<pb16*>(out.data_ptr()),   \
-                     reinterpret_cast<pb16*>(inter.data_ptr()),          \
+                     want_inter ? reinterpret_cast<pb16*>(inter.data_ptr())\
+                                : nullptr,                               \
                      (int)M, N, K, mtiles, ntiles)
 #define LAUNCH_ST(HB)                                                    \
   do { if (glds) LAUNCH(HB, STAGE_GLDS); else LAUNCH(HB, STAGE_PAD); }   \
@@ -375,5 +381,6 @@ std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
   if (bp) LAUNCH_ST(true); else LAUNCH_ST(false);
 #undef LAUNCH_ST
 #undef LAUNCH
-  return {out, inter};
+  if (want_inter) return {out, inter};
+  return {out};
 }

@@ -564,3 +564,22 @@ def test_pairrep_build_parity(ext):
         denom = g2.abs().max().item()
         err = (g1.float() - g2).abs().max().item()
         assert err < 6e-2 * denom + 1e-3, (err, denom)
+
+
+def test_ff1_geglu_inference_no_inter(ext):
+    """Inference path: want_inter=False returns only the gated output
+    (no pre-activation materialization) and matches the training path."""
+    torch.manual_seed(5)
+    x = torch.randn(256, 256, device='cuda', dtype=torch.bfloat16)
+    w = torch.randn(2048, 256, device='cuda', dtype=torch.bfloat16) * 0.1
+    b = torch.randn(2048, device='cuda', dtype=torch.bfloat16)
+    full = ext.ff1_geglu_fwd(x, w, b)
+    lean = ext.ff1_geglu_fwd(x, w, b, -1, False)
+    assert len(full) == 2 and len(lean) == 1
+    assert torch.equal(full[0], lean[0])
+
+    # dispatch-level: no_grad routes to the lean path
+    from alphafold2_amd.ops import dispatch
+    with torch.no_grad():
+        out = dispatch.ff1_geglu(x, w, b)
+    assert torch.equal(out, full[0])
