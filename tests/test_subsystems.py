@@ -452,3 +452,31 @@ def test_enable_tuned_gemm_cpu_noop():
     assert os.path.exists(tuning._CACHE)
     head = open(tuning._CACHE).read(200)
     assert 'Validator' in head and 'GCN_ARCH_NAME' in head
+
+
+@pytest.mark.timeout(300)
+def test_train_end2end_scn_data(tmp_path):
+    import pickle
+    import subprocess
+    import sys
+    rng = np.random.default_rng(2)
+    seqs = ['ACDEFGHIKLMNPQ', 'GHIKLMNPQRSTVW']
+    data = {'train': {
+        'seq': seqs,
+        'crd': [rng.normal(size=(len(s) * 14, 3)).astype(np.float32)
+                for s in seqs],
+        'msk': ['+' * len(s) for s in seqs],
+        'ids': ['a', 'b']}}
+    path = tmp_path / 'scn.pkl'
+    with open(path, 'wb') as f:
+        pickle.dump(data, f)
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, 'train_end2end.py', '--scn-data', str(path),
+         '--batches', '2', '--grad-accum', '1', '--dim', '32',
+         '--depth', '1', '--structure-depth', '1', '--batch-size', '2',
+         '--dtype', 'fp32', '--embedder', 'none',
+         '--checkpoint', str(tmp_path / 'c.pt'), '--save-every', '0'],
+        cwd=root, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert 'loss' in r.stdout

@@ -29,6 +29,9 @@ def parse_args():
     p.add_argument('--msa-depth', type=int, default=32)
     p.add_argument('--batch-size', type=int, default=1)
     p.add_argument('--data', type=str, default=None)
+    p.add_argument('--scn-data', type=str, default=None,
+                   help='sidechainnet-format pickle (the reference '
+                        'end2end data source; CA coords from crd)')
     p.add_argument('--embedder', type=str, default='none',
                    choices=['none', 'fake', 'esm'],
                    help='frozen-LM embedding front-end')
@@ -84,17 +87,25 @@ def main():
         if rank == 0:
             print(f'resumed from {args.checkpoint} at step {start_step}')
 
-    if args.data is not None:
+    if args.scn_data is not None:
+        from alphafold2_amd.data import scn as scn_fmt
+        dls = scn_fmt.load(args.scn_data, batch_size=args.batch_size,
+                           crop_len=min(args.max_len, 256), seed=rank)
+        train_key = next(k for k in dls if k.startswith('train'))
+        dl = dls[train_key]
+    elif args.data is not None:
         from alphafold2_amd.data.trrosetta import TrRosettaDataset
         ds = TrRosettaDataset(args.data, max_seq_len=args.max_len,
                               max_msa_depth=args.msa_depth)
+        dl = torch.utils.data.DataLoader(ds, batch_size=args.batch_size,
+                                         num_workers=0)
     else:
         ds = SyntheticProteinDataset(length=args.batches * args.batch_size,
                                      seq_len=min(args.max_len, 256),
                                      msa_depth=args.msa_depth,
                                      seed=2000 + rank)
-    dl = torch.utils.data.DataLoader(ds, batch_size=args.batch_size,
-                                     num_workers=0)
+        dl = torch.utils.data.DataLoader(ds, batch_size=args.batch_size,
+                                         num_workers=0)
     data_iter = iter(dl)
 
     use_bf16 = args.dtype == 'bf16' and device.type == 'cuda'
@@ -122,10 +133,15 @@ def main():
                 data_iter = iter(dl)
                 batch = next(data_iter)
             seq = batch['seq'].to(device)
-            msa = batch['msa'].to(device)
             mask = batch['mask'].to(device)
-            msa_mask = batch['msa_mask'].to(device)
-            coords = batch['coords'].to(device)
+            if 'msa' in batch:
+                msa = batch['msa'].to(device)
+                msa_mask = batch['msa_mask'].to(device)
+            else:
+                # sidechainnet path: seq-only, sequence as its own MSA row
+                msa = seq[:, None, :]
+                msa_mask = mask[:, None, :]
+            coords = batch.get('ca_coords', batch['coords']).to(device)
             target = get_bucketed_distance_matrix(coords, mask)
 
             sync_ctx = engine.no_sync() if micro < args.grad_accum - 1 \
