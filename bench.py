@@ -210,6 +210,10 @@ def main():
     for _ in range(args.warmup):
         step()
     sync()
+    if rank == 0 and device.type == 'cuda':
+        peak = torch.cuda.max_memory_allocated(device) / 2**30
+        total = torch.cuda.get_device_properties(device).total_memory / 2**30
+        print(f"# peak HBM: {peak:.1f} / {total:.0f} GiB", flush=True)
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
