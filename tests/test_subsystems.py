@@ -480,3 +480,48 @@ def test_train_end2end_scn_data(tmp_path):
         cwd=root, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-2000:]
     assert 'loss' in r.stdout
+
+
+def test_mlm_subset_sampler_quota_invariants():
+    """Rank-based subset sampler: per-row quota = min(ceil(p*allowed),
+    ceil(p*len)), only allowed positions picked, uniform-ish coverage."""
+    from alphafold2_amd.mlm import get_mask_subset_with_prob
+    import math
+    torch.manual_seed(0)
+    b, n, p = 64, 40, 0.15
+    mask = torch.rand(b, n) > 0.3
+    sel = get_mask_subset_with_prob(mask, p)
+    assert not (sel & ~mask).any()          # never picks disallowed
+    budget = math.ceil(p * n)
+    for i in range(b):
+        allowed = int(mask[i].sum())
+        quota = min(math.ceil(p * allowed), budget)
+        assert int(sel[i].sum()) == quota, (i, allowed)
+
+    # repeated draws cover different positions (uniformity smoke)
+    counts = torch.zeros(n)
+    full = torch.ones(1, n, dtype=torch.bool)
+    for _ in range(200):
+        counts += get_mask_subset_with_prob(full, p)[0].float()
+    assert (counts > 0).float().mean() > 0.95
+
+
+def test_trunk_conversion_is_involution():
+    from alphafold2_amd.runtime import convert_trunk_state_dict
+    from alphafold2_amd import Alphafold2
+    sd = Alphafold2(dim=32, depth=1, heads=2, dim_head=16).state_dict()
+    back = convert_trunk_state_dict(
+        convert_trunk_state_dict(sd, to='reversible'), to='standard')
+    assert list(back.keys()) == list(sd.keys())
+
+
+def test_dual_backend_auto_dispatch():
+    import numpy as np
+    from alphafold2_amd.utils import RMSD
+    a = torch.randn(3, 10)
+    b = torch.randn(3, 10)
+    t = RMSD(a, b)
+    n = RMSD(a.numpy(), b.numpy())
+    assert isinstance(t, torch.Tensor)
+    assert isinstance(n, np.ndarray)
+    assert abs(float(t[0]) - float(n[0])) < 1e-5
