@@ -71,6 +71,59 @@ class InvariantPointAttention(nn.Module):
             heads * (scalar_value_dim + pairwise_repr_dim
                      + point_value_dim * (3 + 1)), dim)
 
+    def _fused_forward(self, x, pairwise_repr, rotations, translations,
+                       mask):
+        """Inference-path fused fp32 IPA core (K7): one kernel for
+        logits + softmax + all aggregations (ops/hip/ipacore.hip,
+        HW-verified bit-exact by tools/ipa_probe.hip).  Returns None
+        when the configuration is outside the compiled constants or
+        gradients are required (training keeps eager autograd)."""
+        from ..ops import dispatch as _dispatch
+        ok = (not torch.is_grad_enabled() and x.is_cuda
+              and self.require_pairwise_repr
+              and self.heads == 8 and self.scalar_key_dim == 16
+              and self.scalar_value_dim == 16
+              and self.point_key_dim == 4 and self.point_value_dim == 4
+              and pairwise_repr is not None
+              and pairwise_repr.shape[-1] == 256
+              and (mask is None or bool(mask.all()))
+              and _dispatch.using_hip(x, 'ipa_core_fwd'))
+        if not ok:
+            return None
+        ext = _dispatch._load_ext()
+        b, n, _ = x.shape
+        h = self.heads
+
+        def f32c(t):
+            return t.float().contiguous()
+
+        q_s = f32c(self.to_scalar_q(x).reshape(b, n, h, self.scalar_key_dim))
+        k_s = f32c(self.to_scalar_k(x).reshape(b, n, h, self.scalar_key_dim))
+        v_s = f32c(self.to_scalar_v(x).reshape(b, n, h, self.scalar_value_dim))
+
+        def pts_global(t, p):
+            local = t.reshape(b, n, h, p, 3)
+            g = torch.einsum('b n h p c, b n c d -> b n h p d',
+                             local.float(), rotations.float())
+            return (g + translations.float()[:, :, None, None, :]) \
+                .contiguous()
+
+        q_pg = pts_global(self.to_point_q(x), self.point_key_dim)
+        k_pg = pts_global(self.to_point_k(x), self.point_key_dim)
+        v_pg = pts_global(self.to_point_v(x), self.point_value_dim)
+
+        bias = f32c(self.to_pairwise_attn_bias[0](pairwise_repr)
+                    .permute(0, 3, 1, 2))
+        point_w = F.softplus(self.point_weights).float().contiguous()
+
+        pieces = ext.ipa_core_fwd(
+            q_s, k_s, v_s, q_pg, k_pg, v_pg, bias,
+            f32c(pairwise_repr), f32c(rotations), f32c(translations),
+            point_w, self.scalar_attn_logits_scale,
+            self.pairwise_attn_logits_scale,
+            self.point_attn_logits_scale, self.eps)
+        return self.to_out(pieces)
+
     def forward(self, single_repr, pairwise_repr=None, *, rotations,
                 translations, mask=None):
         x = single_repr
@@ -78,6 +131,11 @@ class InvariantPointAttention(nn.Module):
         h = self.heads
         assert not (self.require_pairwise_repr and not exists(pairwise_repr)), \
             'pairwise representation must be given'
+
+        fused = self._fused_forward(x, pairwise_repr, rotations,
+                                    translations, mask)
+        if fused is not None:
+            return fused
 
         # scalar qkv -> (b, h, n, d)
         def split_heads(t, d):

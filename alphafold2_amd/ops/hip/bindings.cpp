@@ -30,6 +30,12 @@ std::vector<at::Tensor> ff1_geglu_fwd(at::Tensor x, at::Tensor W,
                                       c10::optional<at::Tensor> bias,
                                       long stage, bool want_inter);
 at::Tensor wgrad(at::Tensor dY, at::Tensor X);
+at::Tensor ipa_core_fwd(at::Tensor q_s, at::Tensor k_s, at::Tensor v_s,
+                        at::Tensor q_pg, at::Tensor k_pg, at::Tensor v_pg,
+                        at::Tensor bias, at::Tensor pair, at::Tensor rot,
+                        at::Tensor trans, at::Tensor point_w,
+                        double scale_s, double scale_b, double scale_p,
+                        double eps);
 at::Tensor pairrep_fwd(at::Tensor left, at::Tensor right, at::Tensor emb,
                        at::Tensor rel);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
@@ -75,6 +81,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused pair-rep build: outer sum + rel-pos embedding gather "
         "(gfx950, K13)", py::arg("left"), py::arg("right"), py::arg("emb"),
         py::arg("rel"));
+  m.def("ipa_core_fwd", &ipa_core_fwd,
+        "fused fp32 IPA attention core, inference path (gfx950, K7)");
   m.def("wgrad", &wgrad,
         "split-K weight-gradient GEMM dY^T @ X (gfx950 MFMA, fp32 out)",
         py::arg("dY"), py::arg("X"));

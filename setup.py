@@ -28,6 +28,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "ffgemm.hip"),
         os.path.join(HIP_DIR, "wgrad.hip"),
         os.path.join(HIP_DIR, "pairrep.hip"),
+        os.path.join(HIP_DIR, "ipacore.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
